@@ -1,0 +1,112 @@
+"""waldb client — serialized single-connection query client.
+
+The reference works around node-postgres connection-state bugs with a
+serialized single-connection query queue (ref lib/postgresMgr.js:1990-2172);
+we keep the same discipline: one TCP connection, one outstanding query.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+from typing import Optional
+
+
+class WaldbError(RuntimeError):
+    pass
+
+
+class WaldbClient:
+    def __init__(self, host: str, port: int, connect_timeout_s: float = 5.0,
+                 query_timeout_s: float = 30.0):
+        self.host = host
+        self.port = port
+        self.connect_timeout_s = connect_timeout_s
+        self.query_timeout_s = query_timeout_s
+        self._reader: Optional[asyncio.StreamReader] = None
+        self._writer: Optional[asyncio.StreamWriter] = None
+        self._lock = asyncio.Lock()
+
+    @classmethod
+    def from_url(cls, url: str, **kw) -> "WaldbClient":
+        """Accepts 'tcp://user@host:port/db' (pgUrl shape) or 'host:port'."""
+        text = url
+        if "://" in text:
+            text = text.split("://", 1)[1]
+        if "@" in text:
+            text = text.split("@", 1)[1]
+        text = text.split("/", 1)[0]
+        host, _, port = text.partition(":")
+        return cls(host, int(port), **kw)
+
+    async def _ensure(self) -> None:
+        if self._writer is None or self._writer.is_closing():
+            self._reader, self._writer = await asyncio.wait_for(
+                asyncio.open_connection(self.host, self.port),
+                self.connect_timeout_s)
+
+    async def query(self, req: dict, timeout_s: Optional[float] = None
+                    ) -> dict:
+        async with self._lock:
+            await self._ensure()
+            try:
+                self._writer.write((json.dumps(req) + "\n").encode())
+                await self._writer.drain()
+                line = await asyncio.wait_for(
+                    self._reader.readline(),
+                    timeout_s if timeout_s is not None
+                    else self.query_timeout_s)
+            except (ConnectionError, OSError, asyncio.TimeoutError) as exc:
+                await self._teardown()
+                raise WaldbError("query failed: %r" % (exc,)) from exc
+            if not line:
+                await self._teardown()
+                raise WaldbError("connection closed by server")
+            try:
+                return json.loads(line)
+            except ValueError as exc:
+                await self._teardown()
+                raise WaldbError("bad response") from exc
+
+    async def _teardown(self) -> None:
+        if self._writer is not None:
+            try:
+                self._writer.close()
+            except Exception:
+                pass
+        self._writer = None
+        self._reader = None
+
+    async def close(self) -> None:
+        await self._teardown()
+
+    # ------------------------------------------------------------- helpers
+    async def ping(self, timeout_s: float = 5.0) -> bool:
+        resp = await self.query({"q": "ping"}, timeout_s=timeout_s)
+        return bool(resp.get("ok"))
+
+    async def put(self, key: str, value, timeout_s: Optional[float] = None
+                  ) -> str:
+        resp = await self.query({"q": "put", "k": key, "v": value},
+                                timeout_s=timeout_s)
+        if not resp.get("ok"):
+            raise WaldbError(resp.get("error", "put failed"))
+        return resp["lsn"]
+
+    async def get(self, key: str):
+        resp = await self.query({"q": "get", "k": key})
+        if not resp.get("ok"):
+            raise WaldbError(resp.get("error", "get failed"))
+        return resp.get("v") if resp.get("found") else None
+
+    async def status(self) -> dict:
+        resp = await self.query({"q": "status"})
+        if not resp.get("ok"):
+            raise WaldbError(resp.get("error", "status failed"))
+        return resp
+
+    async def xlog(self) -> str:
+        resp = await self.query({"q": "xlog"})
+        if not resp.get("ok"):
+            raise WaldbError(resp.get("error", "xlog failed"))
+        return resp["lsn"]

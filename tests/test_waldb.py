@@ -1,0 +1,355 @@
+"""waldb engine tests: replication semantics, crash recovery, divergence.
+
+These spawn real waldb subprocesses and kill them with SIGKILL — the only
+way this system ever stops a database (MANATEE-188)."""
+
+import asyncio
+import json
+import os
+import signal
+import socket
+import subprocess
+import sys
+import time
+
+import pytest
+
+from manatee_amd.common import confparser
+from manatee_amd.db.waldb.client import WaldbClient, WaldbError
+from manatee_amd.db.waldb.server import init_data_dir
+
+REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+
+
+def free_port():
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+class Node:
+    def __init__(self, tmp_path, name, port=None):
+        self.name = name
+        self.data_dir = str(tmp_path / name)
+        self.port = port or free_port()
+        self.proc = None
+
+    def init(self):
+        init_data_dir(self.data_dir)
+
+    def write_conf(self, role="primary", upstream=None, sync_name=None,
+                   read_only=False):
+        conf = {
+            "role": role,
+            "listen_ip": "127.0.0.1",
+            "port": str(self.port),
+            "name": self.name,
+            "default_transaction_read_only": "on" if read_only else "off",
+        }
+        if upstream:
+            conf["primary_conninfo"] = "'%s'" % upstream
+        if sync_name:
+            conf["synchronous_standby_names"] = "'%s'" % sync_name
+        confparser.write(os.path.join(self.data_dir, "waldb.conf"), conf)
+
+    def start(self):
+        env = dict(os.environ, PYTHONPATH=REPO)
+        self.proc = subprocess.Popen(
+            [sys.executable, "-m", "manatee_amd.db.waldb.server",
+             "-D", self.data_dir],
+            env=env, stdout=subprocess.DEVNULL, stderr=subprocess.PIPE)
+        # wait for the pid file (readiness)
+        pid_file = os.path.join(self.data_dir, "waldb.pid")
+        deadline = time.time() + 10
+        while not os.path.exists(pid_file) or \
+                os.stat(pid_file).st_size == 0:
+            if self.proc.poll() is not None:
+                raise RuntimeError("waldb died: %s"
+                                   % self.proc.stderr.read().decode())
+            assert time.time() < deadline, "waldb did not start"
+            time.sleep(0.02)
+
+    def kill9(self):
+        self.proc.send_signal(signal.SIGKILL)
+        self.proc.wait()
+        os.unlink(os.path.join(self.data_dir, "waldb.pid"))
+
+    def sighup(self):
+        self.proc.send_signal(signal.SIGHUP)
+
+    def stop(self):
+        if self.proc and self.proc.poll() is None:
+            self.proc.kill()
+            self.proc.wait()
+
+    def client(self):
+        return WaldbClient("127.0.0.1", self.port)
+
+    def promote_trigger(self):
+        open(os.path.join(self.data_dir, "promote"), "w").close()
+
+
+async def wait_async(pred_coro, timeout=10.0, what="condition"):
+    deadline = time.monotonic() + timeout
+    while True:
+        if await pred_coro():
+            return
+        assert time.monotonic() < deadline, "timeout: " + what
+        await asyncio.sleep(0.05)
+
+
+def run(coro):
+    return asyncio.run(asyncio.wait_for(coro, 60))
+
+
+def test_single_node_put_get_and_crash_recovery(tmp_path):
+    n = Node(tmp_path, "n1")
+    n.init()
+    n.write_conf(role="primary")
+    n.start()
+    try:
+        async def phase1():
+            c = n.client()
+            assert await c.ping()
+            lsn1 = await c.put("a", 1)
+            lsn2 = await c.put("b", {"x": [1, 2]})
+            assert lsn2 > lsn1
+            assert await c.get("a") == 1
+            assert await c.get("b") == {"x": [1, 2]}
+            assert await c.get("nope") is None
+            st = await c.status()
+            assert st["role"] == "primary"
+            assert st["current_lsn"] == lsn2
+            await c.close()
+        run(phase1())
+        n.kill9()
+        # corrupt tail: simulate torn write from the dirty kill
+        with open(os.path.join(n.data_dir, "wal.log"), "ab") as f:
+            f.write(b"\x00\x00\x00\x10partial")
+        n.start()
+
+        async def phase2():
+            c = n.client()
+            assert await c.get("a") == 1
+            assert await c.get("b") == {"x": [1, 2]}
+            await c.close()
+        run(phase2())
+    finally:
+        n.stop()
+
+
+def test_sync_replication_gates_commit(tmp_path):
+    prim = Node(tmp_path, "prim")
+    sync = Node(tmp_path, "sync")
+    prim.init()
+    prim.write_conf(role="primary", sync_name="sync")
+    prim.start()
+    try:
+        async def no_standby_blocks():
+            c = n_client = prim.client()
+            with pytest.raises(WaldbError):
+                # no sync standby connected → commit must block
+                await c.put("k", "v", timeout_s=1.0)
+            await c.close()
+        run(no_standby_blocks())
+
+        # bootstrap the sync from the primary's data dir (they must share
+        # the system identifier)
+        import shutil
+        shutil.copytree(prim.data_dir, sync.data_dir,
+                        ignore=shutil.ignore_patterns("waldb.pid",
+                                                      "waldb.conf"))
+        sync.write_conf(role="standby",
+                        upstream="127.0.0.1:%d" % prim.port)
+        sync.start()
+
+        async def with_standby():
+            c = prim.client()
+            sc = sync.client()
+
+            async def sync_streaming():
+                st = await sc.status()
+                return st["upstream_status"] == "streaming"
+            await wait_async(sync_streaming, what="sync streaming")
+            lsn = await c.put("k", "v", timeout_s=5.0)
+
+            async def replicated():
+                return await sc.get("k") == "v"
+            await wait_async(replicated, what="value on sync")
+            st = await c.status()
+            assert len(st["replication"]) == 1
+            r = st["replication"][0]
+            assert r["application_name"] == "sync"
+            assert r["sync_state"] == "sync"
+            assert r["write_lsn"] >= lsn
+            # standby rejects writes
+            with pytest.raises(WaldbError):
+                await sc.put("x", 1)
+            await c.close()
+            await sc.close()
+        run(with_standby())
+    finally:
+        prim.stop()
+        sync.stop()
+
+
+def test_cascading_replication_and_promote_divergence(tmp_path):
+    import shutil
+    prim = Node(tmp_path, "prim")
+    sync = Node(tmp_path, "syncp")
+    asy = Node(tmp_path, "asyncp")
+    prim.init()
+    prim.write_conf(role="primary", sync_name="syncp")
+    prim.start()
+    try:
+        async def seed():
+            c = prim.client()
+            # ONWM-style: temporarily no sync gate
+            prim.write_conf(role="primary")
+            prim.sighup()
+            await asyncio.sleep(0.2)
+            for i in range(50):
+                await c.put("seed%d" % i, i)
+            await c.close()
+        run(seed())
+
+        for node in (sync, asy):
+            shutil.copytree(prim.data_dir, node.data_dir,
+                            ignore=shutil.ignore_patterns("waldb.pid",
+                                                          "waldb.conf"))
+        sync.write_conf(role="standby", upstream="127.0.0.1:%d" % prim.port)
+        asy.write_conf(role="standby", upstream="127.0.0.1:%d" % sync.port)
+        sync.start()
+        asy.start()
+        prim.write_conf(role="primary", sync_name="syncp")
+        prim.sighup()
+
+        async def chain():
+            pc, sc, ac = prim.client(), sync.client(), asy.client()
+
+            async def both_streaming():
+                s1 = await sc.status()
+                s2 = await ac.status()
+                return (s1["upstream_status"] == "streaming"
+                        and s2["upstream_status"] == "streaming")
+            await wait_async(both_streaming, what="cascade streaming")
+            await pc.put("cascade", "yes", timeout_s=5.0)
+
+            async def reached_async():
+                return await ac.get("cascade") == "yes"
+            await wait_async(reached_async, what="value cascaded to async")
+            # the async's row appears on the SYNC (cascading), not primary
+            st = await sc.status()
+            assert [r["application_name"]
+                    for r in st["replication"]] == ["asyncp"]
+            await pc.close()
+            await sc.close()
+            await ac.close()
+        run(chain())
+
+        # ---- failover: kill primary; promote the sync (timeline bump) ----
+        prim.kill9()
+        sync.kill9()
+        sync.promote_trigger()
+        sync.write_conf(role="primary")
+        sync.start()
+        asy.write_conf(role="standby", upstream="127.0.0.1:%d" % sync.port)
+        asy.sighup()
+
+        async def promoted():
+            sc, ac = sync.client(), asy.client()
+            st = await sc.status()
+            assert st["role"] == "primary"
+            assert st["timeline"] == 2
+            await sc.put("after-failover", 1)
+
+            async def follows():
+                s = await ac.status()
+                return (s["upstream_status"] == "streaming"
+                        and s["timeline"] == 2)
+            await wait_async(follows, what="async follows new timeline")
+
+            async def got():
+                return await ac.get("after-failover") == 1
+            await wait_async(got, what="post-failover write cascaded")
+            await sc.close()
+            await ac.close()
+        run(promoted())
+
+        # ---- divergence: old primary writes past the switch point, then
+        # tries to follow the new primary → must be refused ----
+        prim.write_conf(role="primary")
+        prim.start()
+
+        async def diverge():
+            pc = prim.client()
+            await pc.put("diverged-write", 1)
+            await pc.close()
+        run(diverge())
+        prim.kill9()
+        prim.write_conf(role="standby", upstream="127.0.0.1:%d" % sync.port)
+        prim.start()
+
+        async def refused():
+            pc = prim.client()
+
+            async def diverged():
+                st = await pc.status()
+                return st["upstream_status"] == "diverged"
+            await wait_async(diverged, what="divergence detected")
+            await pc.close()
+        run(refused())
+    finally:
+        prim.stop()
+        sync.stop()
+        asy.stop()
+
+
+def test_acked_writes_survive_primary_kill9(tmp_path):
+    """The north-star durability property: every acknowledged write is on
+    the sync standby after kill -9 of the primary (BASELINE.json)."""
+    import shutil
+    prim = Node(tmp_path, "prim")
+    sync = Node(tmp_path, "s1")
+    prim.init()
+    prim.write_conf(role="primary")
+    prim.start()
+    try:
+        run(prim.client().ping())
+        shutil.copytree(prim.data_dir, sync.data_dir,
+                        ignore=shutil.ignore_patterns("waldb.pid",
+                                                      "waldb.conf"))
+        sync.write_conf(role="standby", upstream="127.0.0.1:%d" % prim.port)
+        sync.start()
+        prim.write_conf(role="primary", sync_name="s1")
+        prim.sighup()
+
+        acked = []
+
+        async def load():
+            c = prim.client()
+            sc = sync.client()
+
+            async def streaming():
+                return (await sc.status())["upstream_status"] == "streaming"
+            await wait_async(streaming, what="sync streaming")
+            for i in range(200):
+                lsn = await c.put("key%d" % i, i, timeout_s=5.0)
+                acked.append(("key%d" % i, i, lsn))
+            await c.close()
+            await sc.close()
+        run(load())
+        prim.kill9()
+
+        async def verify():
+            sc = sync.client()
+            for k, v, _lsn in acked:
+                assert await sc.get(k) == v, \
+                    "acknowledged write %s lost after primary kill -9" % k
+            await sc.close()
+        run(verify())
+    finally:
+        prim.stop()
+        sync.stop()
