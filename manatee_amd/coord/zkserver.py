@@ -574,3 +574,27 @@ async def run_standalone(host: str, port: int, log: Logger,
             await asyncio.sleep(3600)
     finally:
         await srv.stop()
+
+
+def main(argv=None) -> int:
+    import argparse
+
+    from ..common.logging import Logger as _Logger, level_from_verbosity
+    ap = argparse.ArgumentParser(prog="manatee-zk")
+    ap.add_argument("-H", "--host", default="127.0.0.1")
+    ap.add_argument("-p", "--port", type=int, default=2181)
+    ap.add_argument("-j", "--journal", default=None)
+    ap.add_argument("-v", "--verbose", action="count", default=0)
+    ns = ap.parse_args(argv)
+    log = _Logger("manatee-zk", level=level_from_verbosity(ns.verbose))
+    try:
+        asyncio.run(run_standalone(ns.host, ns.port, log,
+                                   journal_path=ns.journal))
+    except KeyboardInterrupt:
+        pass
+    return 0
+
+
+if __name__ == "__main__":
+    import sys
+    sys.exit(main())

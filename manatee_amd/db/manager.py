@@ -1,0 +1,427 @@
+"""DbManager — the database lifecycle manager (PostgresMgr equivalent).
+
+Clean-room re-implementation of ``lib/postgresMgr.js``: owns the database
+child process and performs the role transitions the FSM requests.
+
+Preserved behaviors (reference citations inline):
+
+- FSM-facing contract: events ``init {setup, online}``, ``healthy``,
+  ``unhealthy``, fatal ``error``; methods ``reconfigure(cfg)``, ``stop()``,
+  ``get_xlog_location()`` (ref :401-421, :686-928);
+- transitions are serialized by a guard — one at a time (ref
+  ``_transitioning`` :690-692, 850-852), and the background
+  wait-for-standby task is cancellable (ref ``_transitionFunc``
+  :379-385, 1123-1131);
+- ``_primary``: prepare data dir (initdb analogue) → write conf
+  **read-only** → restart → snapshot → background wait until the named
+  sync standby is caught up, then set ``synchronous_standby_names`` +
+  reload and open writes (ref :1115-1184, 1037-1105, 2390-2556);
+- ``_update_standby``: a primary whose downstream changed only rewrites
+  the sync name + SIGHUP — commits then block until the new sync catches
+  up (ref :1195-1260);
+- ``_standby``: stop → write upstream conf → restart; on failure or on WAL
+  divergence, full restore from the restorePeer (always the primary — the
+  back-pressure rule, ref :1019-1029, 1282-1460);
+- stop is an escalating dirty kill (SIGINT → SIGQUIT → SIGKILL, one
+  ops-timeout per step) and NEVER a clean shutdown — avoiding xlog
+  divergence (ref :1484-1541, MANATEE-188);
+- health: periodic ping with interval/timeout semantics
+  (ref :1550-1626, 2373-2383); an unexpected child exit emits a fatal
+  ``error`` event (ref :1711-1753).
+"""
+
+from __future__ import annotations
+
+import asyncio
+import os
+import signal
+import time
+from typing import Any, Callable, Dict, List, Optional
+
+from ..backup.restore import RestoreClient
+from ..common import lsn as lsnmod
+from ..common import procutil
+from ..common.logging import Logger, null_logger
+from ..storage.provider import SnapshotStore
+from .engine import Engine, peer_id_from_urls
+
+
+class DbManager:
+    def __init__(self, *, engine: Engine, store: SnapshotStore,
+                 ip: str,
+                 health_interval_s: float = 1.0,
+                 health_timeout_s: float = 5.0,
+                 ops_timeout_s: float = 60.0,
+                 replication_timeout_s: float = 60.0,
+                 repl_poll_s: float = 0.2,
+                 one_node_write_mode: bool = False,
+                 log: Optional[Logger] = None):
+        self.engine = engine
+        self.store = store
+        self.ip = ip
+        self.health_interval_s = health_interval_s
+        self.health_timeout_s = health_timeout_s
+        self.ops_timeout_s = ops_timeout_s
+        self.replication_timeout_s = replication_timeout_s
+        self.repl_poll_s = repl_poll_s
+        self.one_node_write_mode = one_node_write_mode
+        self.log = (log or null_logger()).child(component="DbManager")
+
+        self.healthy = False
+        self.online = False
+        self.writable = False
+        self._proc: Optional[asyncio.subprocess.Process] = None
+        self._proc_monitor: Optional[asyncio.Task] = None
+        self._expect_exit = False
+        self._lock = asyncio.Lock()          # _transitioning guard
+        self._transition_task: Optional[asyncio.Task] = None
+        self._health_task: Optional[asyncio.Task] = None
+        self._listeners: Dict[str, List[Callable]] = {}
+        self._applied: Optional[dict] = None
+        self.restore_client: Optional[RestoreClient] = None
+        self._closing = False
+
+    # --------------------------------------------------------------- events
+    def on(self, event: str, cb: Callable) -> None:
+        self._listeners.setdefault(event, []).append(cb)
+
+    def _emit(self, event: str, *args: Any) -> None:
+        for cb in self._listeners.get(event, []):
+            try:
+                cb(*args)
+            except Exception as exc:
+                self.log.error("listener error", event=event, err=exc)
+
+    # ------------------------------------------------------------ lifecycle
+    async def start(self) -> None:
+        """Probe setup/online and emit ``init`` (ref :401-421)."""
+        self._health_task = asyncio.get_running_loop().create_task(
+            self._health_loop())
+        setup = self.engine.initialized()
+        online = await self.engine.ping(timeout_s=1.0) if setup else False
+        self.online = online
+        self.healthy = online
+        self._emit("init", {"setup": setup, "online": online})
+
+    async def close(self) -> None:
+        self._closing = True
+        self._cancel_transition()
+        if self._health_task is not None:
+            self._health_task.cancel()
+        await self._stop_db()
+        await self.engine.close()
+
+    # --------------------------------------------------------------- health
+    async def _health_loop(self) -> None:
+        self._last_health_ok = time.monotonic()
+        while not self._closing:
+            await asyncio.sleep(self.health_interval_s)
+            if self._proc is None:
+                self._last_health_ok = time.monotonic()
+                continue
+            ok = await self.engine.ping(timeout_s=self.health_timeout_s)
+            now = time.monotonic()
+            if ok:
+                self._last_health_ok = now
+                if not self.healthy:
+                    self.healthy = True
+                    self.online = True
+                    self.log.info("database is healthy again")
+                    self._emit("healthy")
+            elif self.healthy and \
+                    (now - self._last_health_ok) > self.health_timeout_s:
+                self.healthy = False
+                self.log.warn("database unhealthy",
+                              since_s=now - self._last_health_ok)
+                self._emit("unhealthy")
+
+    # ------------------------------------------------------- process control
+    async def _start_db(self) -> None:
+        if self._proc is not None:
+            return
+        self._expect_exit = False
+        argv = self.engine.spawn_argv()
+        self.log.info("starting database", argv=argv)
+        self._proc = await asyncio.create_subprocess_exec(
+            *argv, stdout=asyncio.subprocess.DEVNULL,
+            stderr=asyncio.subprocess.DEVNULL,
+            start_new_session=True)
+        self._proc_monitor = asyncio.get_running_loop().create_task(
+            self._monitor_proc(self._proc))
+        # poll until the db answers (ref _start 1 Hz poll :1760-1794;
+        # we poll faster to shrink failover time)
+        deadline = time.monotonic() + self.ops_timeout_s
+        while True:
+            if self._proc is None or self._proc.returncode is not None:
+                raise RuntimeError("database exited during startup")
+            if await self.engine.ping(timeout_s=1.0):
+                break
+            if time.monotonic() > deadline:
+                raise RuntimeError("database did not become ready in %ss"
+                                   % self.ops_timeout_s)
+            await asyncio.sleep(0.05)
+        self.online = True
+        self.healthy = True
+        self._last_health_ok = time.monotonic()
+
+    async def _monitor_proc(self, proc: asyncio.subprocess.Process) -> None:
+        rc = await proc.wait()
+        if self._proc is proc:
+            self._proc = None
+        if not self._expect_exit and not self._closing:
+            # unexpected death → fatal error event (ref :1711-1753)
+            self.log.error("database exited unexpectedly", rc=rc)
+            self.online = False
+            self.healthy = False
+            self.writable = False
+            self._applied = None
+            self._emit("error", RuntimeError("db exited rc=%s" % rc))
+
+    async def _stop_db(self) -> None:
+        proc = self._proc
+        if proc is None:
+            return
+        self._expect_exit = True
+        self._proc = None
+        self.online = False
+        self.healthy = False
+        self.writable = False
+        if proc.returncode is None:
+            try:
+                await procutil.kill_escalate(proc.pid, self.ops_timeout_s,
+                                             pgid=True)
+            except (procutil.ExecError, ProcessLookupError):
+                pass
+        try:
+            await asyncio.wait_for(proc.wait(), self.ops_timeout_s)
+        except asyncio.TimeoutError:
+            pass
+        self.log.info("database stopped")
+
+    async def _restart_db(self) -> None:
+        await self._stop_db()
+        await self._start_db()
+
+    # ----------------------------------------------------------- transitions
+    def _cancel_transition(self) -> None:
+        if self._transition_task is not None:
+            self._transition_task.cancel()
+            self._transition_task = None
+
+    async def reconfigure(self, cfg: dict) -> None:
+        """Apply a role configuration from the FSM (contract ref :758-867)."""
+        async with self._lock:
+            self._cancel_transition()
+            role = cfg.get("role")
+            if role == "none":
+                await self._stop_db()
+                self._applied = cfg
+                return
+            if role == "primary":
+                prev = self._applied
+                if (prev and prev.get("role") == "primary"
+                        and self._proc is not None):
+                    await self._update_standby(cfg)
+                else:
+                    await self._primary(cfg)
+            else:
+                await self._standby(cfg)
+            self._applied = cfg
+
+    # ---------------------------------------------------------- primary path
+    async def _primary(self, cfg: dict) -> None:
+        """Full primary transition (ref _primary :1115-1184)."""
+        downstream = cfg.get("downstream")
+        self.log.info("transitioning to primary",
+                      downstream=(downstream or {}).get("pgUrl"))
+        await self._stop_db()
+        if not self.engine.initialized():
+            await self.store.ensure()
+            await self.engine.init_datadir()
+        was_standby = self.engine.current_conf_role() == "standby"
+        onwm = self.one_node_write_mode or downstream is None
+        # start read-only unless ONWM (ref :1145-1154); sync names are set
+        # only after the standby has caught up (ref :1077-1085)
+        self.engine.write_conf("primary", read_only=not onwm)
+        if was_standby:
+            self.engine.write_promote_trigger()
+        await self._restart_db()
+        self.writable = bool(onwm)
+        try:
+            await self.store.snapshot()   # for future bootstraps (ref :1158)
+        except Exception as exc:
+            self.log.warn("post-transition snapshot failed", err=exc)
+        if not onwm:
+            self._transition_task = asyncio.get_running_loop().create_task(
+                self._wait_for_standby(downstream))
+
+    async def _update_standby(self, cfg: dict) -> None:
+        """Downstream swap on a running primary: conf + SIGHUP only
+        (ref _updateStandby :1195-1260).  Commits block until the new sync
+        catches up; we additionally gate writes read-only until then so
+        the writable flag is accurate."""
+        downstream = cfg.get("downstream")
+        onwm = self.one_node_write_mode or downstream is None
+        self.log.info("updating standby config on running primary",
+                      downstream=(downstream or {}).get("pgUrl"))
+        if onwm:
+            self.engine.write_conf("primary", read_only=False)
+            self._reload_db()
+            self.writable = True
+            return
+        self.engine.write_conf("primary", read_only=True)
+        self._reload_db()
+        self.writable = False
+        self._transition_task = asyncio.get_running_loop().create_task(
+            self._wait_for_standby(downstream))
+
+    def _reload_db(self) -> None:
+        if self._proc is not None and self._proc.returncode is None:
+            try:
+                os.kill(self._proc.pid, signal.SIGHUP)
+            except ProcessLookupError:
+                pass
+
+    async def _wait_for_standby(self, downstream: dict) -> None:
+        """Poll replication status until the sync standby has caught up,
+        then set synchronous_standby_names + open writes
+        (ref _waitForStandby :1037-1105 + _checkRepl :2390-2475)."""
+        standby_name = peer_id_from_urls(downstream["pgUrl"],
+                                         downstream["backupUrl"])
+        deadline = time.monotonic() + self.replication_timeout_s
+        last_progress: Optional[str] = None
+        try:
+            while True:
+                try:
+                    repl = await self.engine.check_repl(standby_name)
+                except Exception:
+                    repl = {"connected": False, "caught_up": False}
+                if repl.get("caught_up"):
+                    break
+                # forward progress resets the timeout (ref :2452-2460)
+                flush = repl.get("write_lsn")
+                if flush is not None and flush != last_progress:
+                    last_progress = flush
+                    deadline = time.monotonic() + self.replication_timeout_s
+                if time.monotonic() > deadline:
+                    self.log.error(
+                        "standby did not catch up within the replication "
+                        "timeout; shard stays read-only",
+                        standby=standby_name)
+                    return
+                await asyncio.sleep(self.repl_poll_s)
+            self.engine.write_conf("primary", sync_name=standby_name,
+                                   read_only=False)
+            self._reload_db()
+            self.writable = True
+            self.log.info("sync standby caught up; writes enabled",
+                          standby=standby_name)
+            self._emit("writable")
+        except asyncio.CancelledError:
+            raise
+
+    # ---------------------------------------------------------- standby path
+    async def _standby(self, cfg: dict) -> None:
+        """Sync/async transition (ref _standby :1282-1460)."""
+        upstream = cfg["upstream"]
+        restore_peer = cfg.get("restorePeer") or upstream
+        role = cfg.get("role", "sync")
+        self.log.info("transitioning to standby", role=role,
+                      upstream=upstream.get("pgUrl"))
+        await self._stop_db()
+        self.writable = False
+        need_restore = not self.engine.initialized()
+        if not need_restore:
+            self.engine.write_conf("standby",
+                                   upstream_url=upstream["pgUrl"])
+            try:
+                await self._start_db()
+            except Exception as exc:
+                self.log.warn("standby failed to start; falling back to "
+                              "restore", err=exc)
+                need_restore = True
+        if not need_restore:
+            # wait briefly for streaming; divergence ⇒ restore
+            verdict = await self._await_streaming(timeout_s=15.0)
+            if verdict == "diverged":
+                self.log.warn("WAL diverged from upstream; restoring")
+                need_restore = True
+            elif verdict == "disconnected":
+                self.log.warn("upstream not reachable yet; staying up and "
+                              "retrying in background")
+        if need_restore:
+            await self._restore_and_start(upstream, restore_peer)
+
+    async def _await_streaming(self, timeout_s: float) -> str:
+        deadline = time.monotonic() + timeout_s
+        while time.monotonic() < deadline:
+            try:
+                status = await self.engine.status()
+            except Exception:
+                await asyncio.sleep(0.2)
+                continue
+            ustat = status.get("upstream_status")
+            if ustat == "streaming":
+                return "streaming"
+            if ustat == "diverged":
+                return "diverged"
+            await asyncio.sleep(0.2)
+        return "disconnected"
+
+    async def _restore_and_start(self, upstream: dict,
+                                 restore_peer: dict) -> None:
+        """Full restore from the restore peer's backup server, then start
+        as a standby (ref zfsClient.restore → _standby retry :1375-1448)."""
+        await self._stop_db()
+        self.restore_client = RestoreClient(self.store, self.ip,
+                                            log=self.log)
+        # only data worth keeping is isolated; an empty/uninitialized
+        # dataset is simply replaced.  Restore is retried with backoff —
+        # the restore peer's backup service may still be coming up
+        # (ref restore retry budget, lib/adm.js:71,1613-1617)
+        isolate = self.engine.initialized()
+        attempt = 0
+        while True:
+            attempt += 1
+            try:
+                await self.restore_client.restore(
+                    restore_peer["backupUrl"], isolate=isolate)
+                break
+            except asyncio.CancelledError:
+                raise
+            except Exception as exc:
+                if attempt >= 5:
+                    raise
+                self.log.warn("restore attempt failed; retrying",
+                              attempt=attempt, err=exc)
+                isolate = False  # already isolated (or nothing to keep)
+                await asyncio.sleep(min(2.0 ** attempt * 0.25, 5.0))
+        self.engine.post_restore_fixup()
+        self.engine.write_conf("standby", upstream_url=upstream["pgUrl"])
+        await self._start_db()
+        verdict = await self._await_streaming(timeout_s=30.0)
+        if verdict != "streaming":
+            self.log.error("standby still not streaming after restore",
+                           verdict=verdict)
+
+    # --------------------------------------------------------------- queries
+    async def get_xlog_location(self) -> str:
+        """Current WAL position (role-aware, ref getXLogLocation :878-899)."""
+        return await self.engine.xlog()
+
+    async def stop(self) -> None:
+        async with self._lock:
+            self._cancel_transition()
+            await self._stop_db()
+
+    def status(self) -> dict:
+        return {
+            "engine": self.engine.name,
+            "online": self.online,
+            "healthy": self.healthy,
+            "writable": self.writable,
+            "role": (self._applied or {}).get("role"),
+            "appliedConfig": self._applied,
+            "restore": (self.restore_client.restore_object.as_dict()
+                        if self.restore_client else None),
+        }

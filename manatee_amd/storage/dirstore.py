@@ -54,10 +54,16 @@ class DirStore(SnapshotStore):
         name = name or snapshot_name_now()
         await self.ensure()
         tmp = self._snap_path(name) + ".partial"
-        # tar the live tree; -C so paths inside are relative
-        await procutil.run_async(
+        # tar the live tree; -C so paths inside are relative.  rc 1 is
+        # tar's "file changed as we read it" warning — expected while the
+        # database appends to its WAL, and harmless: any WAL prefix is a
+        # valid restore point (torn tails are truncated on replay).
+        res = await procutil.run_async(
             ["tar", "-cf", tmp, "-C", self.live, "."],
-            env=procutil.SCRUBBED_ENV, timeout=600)
+            env=procutil.SCRUBBED_ENV, timeout=600, check=False)
+        if res.returncode not in (0, 1):
+            raise procutil.ExecError(res.argv, res.returncode, res.stdout,
+                                     res.stderr)
         os.replace(tmp, self._snap_path(name))
         self.log.debug("snapshot created", snapshot=name)
         return name

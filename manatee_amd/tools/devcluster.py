@@ -1,0 +1,342 @@
+"""DevCluster — build and drive a multi-peer shard on one host.
+
+The equivalent of ``tools/mkdevsitters`` + ``test/testManatee.js``: creates
+per-peer directories and configs (ports stepped +10 per peer like the
+reference, ref docs/working-on-manatee.md:179-197), spawns the embedded ZK
+server, and one sitter + backupserver (+ optional snapshotter) pair per
+peer as real subprocesses.  Used by the integration tests, ``bench.py``
+and ``tools/mkdevsitters``.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import os
+import signal
+import socket
+import subprocess
+import sys
+import time
+from typing import Dict, List, Optional
+
+from ..common.httpd import http_request
+from ..db.waldb.client import WaldbClient
+
+REPO_ROOT = os.path.dirname(os.path.dirname(
+    os.path.dirname(os.path.abspath(__file__))))
+
+
+def free_port() -> int:
+    s = socket.socket()
+    s.bind(("127.0.0.1", 0))
+    port = s.getsockname()[1]
+    s.close()
+    return port
+
+
+def _ephemeral_low() -> int:
+    """Lower bound of the kernel's ephemeral port range — peer ports must
+    stay BELOW it or random client sockets will collide with them."""
+    try:
+        with open("/proc/sys/net/ipv4/ip_local_port_range") as f:
+            return int(f.read().split()[0])
+    except (OSError, ValueError, IndexError):
+        return 32768
+
+
+_EPHEMERAL_LOW = _ephemeral_low()
+
+
+class DevPeer:
+    def __init__(self, cluster: "DevCluster", index: int, base_port: int):
+        self.cluster = cluster
+        self.index = index
+        self.ip = cluster.ip
+        self.pg_port = base_port
+        self.status_port = base_port + 1
+        self.backup_port = base_port + 2
+        self.id = "%s:%d:%d" % (self.ip, self.pg_port, self.backup_port)
+        self.dir = os.path.join(cluster.base_dir, "peer%d" % index)
+        self.store_dir = os.path.join(self.dir, "store")
+        self.sitter_proc: Optional[subprocess.Popen] = None
+        self.backup_proc: Optional[subprocess.Popen] = None
+
+    # ------------------------------------------------------------- configs
+    def sitter_config(self) -> dict:
+        c = self.cluster
+        return {
+            "ip": self.ip,
+            "postgresPort": self.pg_port,
+            "backupPort": self.backup_port,
+            "shardPath": c.shard_path,
+            "zoneId": "peer%d" % self.index,
+            "zkCfg": {
+                "connStr": c.zk_conn_str,
+                "opts": {"sessionTimeout": c.session_timeout_ms},
+            },
+            "postgresMgrCfg": {
+                "engine": c.engine,
+                "storageCfg": {"provider": "dir",
+                               "mountpoint": os.path.join(self.store_dir)},
+                "healthChkInterval": c.health_interval_ms,
+                "healthChkTimeout": c.health_timeout_ms,
+                "opsTimeout": c.ops_timeout_ms,
+                "replicationTimeout": c.replication_timeout_ms,
+                "tickInterval": c.tick_interval_ms,
+                "oneNodeWriteMode": c.singleton,
+            },
+        }
+
+    def backupserver_config(self) -> dict:
+        return {
+            "ip": self.ip,
+            "backupServerCfg": {"port": self.backup_port},
+            "backupSenderCfg": {
+                "storageCfg": {"provider": "dir",
+                               "mountpoint": os.path.join(self.store_dir)},
+            },
+        }
+
+    def write_configs(self) -> None:
+        os.makedirs(self.dir, exist_ok=True)
+        with open(os.path.join(self.dir, "sitter.json"), "w") as f:
+            json.dump(self.sitter_config(), f, indent=2)
+        with open(os.path.join(self.dir, "backupserver.json"), "w") as f:
+            json.dump(self.backupserver_config(), f, indent=2)
+
+    # ------------------------------------------------------------- control
+    def _spawn(self, module: str, config: str, logname: str
+               ) -> subprocess.Popen:
+        env = dict(os.environ)
+        env["PYTHONPATH"] = REPO_ROOT + os.pathsep + \
+            env.get("PYTHONPATH", "")
+        logf = open(os.path.join(self.dir, logname), "a")
+        return subprocess.Popen(
+            [sys.executable, "-m", module, "-f",
+             os.path.join(self.dir, config), "-v",
+             "--log-file", os.path.join(self.dir, logname + ".json")],
+            env=env, stdout=logf, stderr=logf, start_new_session=True)
+
+    def start(self) -> None:
+        self.write_configs()
+        self.backup_proc = self._spawn("manatee_amd.daemons.backupserver",
+                                       "backupserver.json", "backupserver.log")
+        self.sitter_proc = self._spawn("manatee_amd.daemons.sitter",
+                                       "sitter.json", "sitter.log")
+
+    def db_pid(self) -> Optional[int]:
+        pid_file = os.path.join(self.store_dir, "live", "data", "waldb.pid")
+        try:
+            with open(pid_file) as f:
+                return int(f.read().split()[0])
+        except (OSError, ValueError, IndexError):
+            return None
+
+    def kill9(self) -> None:
+        """SIGKILL the whole peer: sitter process group + db child +
+        backupserver — the integ-test failure mode
+        (ref test/integ.test.js primaryDeath et al)."""
+        db_pid = self.db_pid()
+        for proc in (self.sitter_proc, self.backup_proc):
+            if proc is not None and proc.poll() is None:
+                try:
+                    os.killpg(proc.pid, signal.SIGKILL)
+                except ProcessLookupError:
+                    pass
+                proc.wait()
+        if db_pid is not None:
+            try:
+                os.killpg(db_pid, signal.SIGKILL)
+            except (ProcessLookupError, PermissionError):
+                try:
+                    os.kill(db_pid, signal.SIGKILL)
+                except ProcessLookupError:
+                    pass
+        self.sitter_proc = None
+        self.backup_proc = None
+
+    def kill_db_only(self) -> None:
+        """SIGKILL only the database child (the sitter must notice and
+        restart it)."""
+        db_pid = self.db_pid()
+        if db_pid is not None:
+            try:
+                os.kill(db_pid, signal.SIGKILL)
+            except ProcessLookupError:
+                pass
+
+    def stop(self) -> None:
+        self.kill9()
+
+    def alive(self) -> bool:
+        return self.sitter_proc is not None and \
+            self.sitter_proc.poll() is None
+
+    # -------------------------------------------------------------- clients
+    def db_client(self) -> WaldbClient:
+        return WaldbClient(self.ip, self.pg_port)
+
+    async def http_status(self, path: str = "/state"):
+        status, body = await http_request(
+            "http://%s:%d%s" % (self.ip, self.status_port, path),
+            timeout_s=5)
+        return status, body
+
+
+class DevCluster:
+    def __init__(self, base_dir: str, n_peers: int = 3,
+                 ip: str = "127.0.0.1",
+                 engine: str = "waldb",
+                 shard_name: str = "1.dev",
+                 session_timeout_ms: int = 2000,
+                 health_interval_ms: int = 500,
+                 health_timeout_ms: int = 3000,
+                 ops_timeout_ms: int = 30000,
+                 replication_timeout_ms: int = 30000,
+                 tick_interval_ms: int = 250,
+                 singleton: bool = False,
+                 base_port: Optional[int] = None):
+        self.base_dir = os.path.abspath(base_dir)
+        self.ip = ip
+        self.engine = engine
+        self.shard_path = "/manatee/" + shard_name
+        self.session_timeout_ms = session_timeout_ms
+        self.health_interval_ms = health_interval_ms
+        self.health_timeout_ms = health_timeout_ms
+        self.ops_timeout_ms = ops_timeout_ms
+        self.replication_timeout_ms = replication_timeout_ms
+        self.tick_interval_ms = tick_interval_ms
+        self.singleton = singleton
+        self.zk_port = 0  # assigned below the ephemeral range in __init__
+        self.zk_conn_str = ""
+        self.zk_proc: Optional[subprocess.Popen] = None
+        self.peers: List[DevPeer] = []
+        # pick a base below the ephemeral range so client sockets can never
+        # collide with peer listen ports
+        lo = max(10000, _EPHEMERAL_LOW - 22000)
+        self._next_base_port = base_port or \
+            (lo + (os.getpid() * 131) % 18000) // 10 * 10
+        zk_peer = self.add_peer_config()   # reserve a port block for ZK
+        self.peers.clear()
+        self.zk_port = zk_peer.pg_port
+        self.zk_conn_str = "%s:%d" % (ip, self.zk_port)
+        for _ in range(n_peers):
+            self.add_peer_config()
+
+    def add_peer_config(self) -> DevPeer:
+        # ports stepped +10 per peer (ref mkdevsitters port scheme)
+        while True:
+            base = self._next_base_port
+            self._next_base_port += 10
+            if base + 10 >= _EPHEMERAL_LOW:
+                base = self._next_base_port = 10000
+                self._next_base_port += 10
+            try:
+                for off in (0, 1, 2):
+                    probe = socket.socket()
+                    probe.setsockopt(socket.SOL_SOCKET,
+                                     socket.SO_REUSEADDR, 1)
+                    probe.bind((self.ip, base + off))
+                    probe.close()
+                break
+            except OSError:
+                continue
+        peer = DevPeer(self, len(self.peers), base)
+        self.peers.append(peer)
+        return peer
+
+    # -------------------------------------------------------------- control
+    def start_zk(self) -> None:
+        env = dict(os.environ)
+        env["PYTHONPATH"] = REPO_ROOT + os.pathsep + \
+            env.get("PYTHONPATH", "")
+        os.makedirs(self.base_dir, exist_ok=True)
+        logf = open(os.path.join(self.base_dir, "zk.log"), "a")
+        self.zk_proc = subprocess.Popen(
+            [sys.executable, "-m", "manatee_amd.coord.zkserver",
+             "-H", self.ip, "-p", str(self.zk_port),
+             "-j", os.path.join(self.base_dir, "zk-journal.jsonl")],
+            env=env, stdout=logf, stderr=logf, start_new_session=True)
+
+    async def wait_zk(self, timeout_s: float = 15.0) -> None:
+        deadline = time.monotonic() + timeout_s
+        while True:
+            try:
+                r, w = await asyncio.open_connection(self.ip, self.zk_port)
+                w.close()
+                return
+            except OSError:
+                if time.monotonic() > deadline:
+                    raise RuntimeError("zk server did not start")
+                await asyncio.sleep(0.05)
+
+    async def start(self, peers: Optional[List[int]] = None) -> None:
+        self.start_zk()
+        await self.wait_zk()
+        for i, peer in enumerate(self.peers):
+            if peers is None or i in peers:
+                peer.start()
+
+    def stop(self) -> None:
+        for peer in self.peers:
+            peer.stop()
+        if self.zk_proc is not None and self.zk_proc.poll() is None:
+            try:
+                os.killpg(self.zk_proc.pid, signal.SIGKILL)
+            except ProcessLookupError:
+                pass
+            self.zk_proc.wait()
+            self.zk_proc = None
+
+    # ------------------------------------------------------------ inspection
+    async def cluster_state(self) -> Optional[dict]:
+        from ..coord.zkclient import ZkClient
+        from ..coord import jute
+        cli = ZkClient(self.zk_conn_str, session_timeout_ms=5000)
+        try:
+            await cli.connect(timeout_s=5)
+            data, _ = await cli.get_data(self.shard_path + "/state")
+            return json.loads(data)
+        except (jute.ZkError, asyncio.TimeoutError, OSError):
+            return None
+        finally:
+            await cli.close()
+
+    async def wait_cluster(self, pred, timeout_s: float = 30.0,
+                           what: str = "cluster state"):
+        deadline = time.monotonic() + timeout_s
+        while True:
+            s = await self.cluster_state()
+            if s is not None and pred(s):
+                return s
+            if time.monotonic() > deadline:
+                raise AssertionError("timeout waiting for %s; last=%r"
+                                     % (what, s))
+            await asyncio.sleep(0.1)
+
+    def peer_by_id(self, peer_id: str) -> DevPeer:
+        for p in self.peers:
+            if p.id == peer_id:
+                return p
+        raise KeyError(peer_id)
+
+    async def wait_writable(self, timeout_s: float = 60.0) -> DevPeer:
+        """Wait until the cluster primary accepts a write; returns it."""
+        deadline = time.monotonic() + timeout_s
+        last_err = None
+        while time.monotonic() < deadline:
+            s = await self.cluster_state()
+            if s is not None:
+                prim = self.peer_by_id(s["primary"]["id"])
+                cli = prim.db_client()
+                try:
+                    await cli.put("__writable_probe__", time.time(),
+                                  timeout_s=1.0)
+                    await cli.close()
+                    return prim
+                except Exception as exc:
+                    last_err = exc
+                    await cli.close()
+            await asyncio.sleep(0.1)
+        raise AssertionError("cluster never became writable: %r" % last_err)
