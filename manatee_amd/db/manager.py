@@ -341,16 +341,47 @@ class DbManager:
                               "restore", err=exc)
                 need_restore = True
         if not need_restore:
-            # wait briefly for streaming; divergence ⇒ restore
-            verdict = await self._await_streaming(timeout_s=15.0)
+            # wait briefly for streaming; divergence ⇒ restore.  The wait is
+            # short — a long wait here would stall the FSM's event loop
+            # during failover while the new primary restarts.
+            verdict = await self._await_streaming(timeout_s=3.0)
             if verdict == "diverged":
                 self.log.warn("WAL diverged from upstream; restoring")
                 need_restore = True
             elif verdict == "disconnected":
-                self.log.warn("upstream not reachable yet; staying up and "
-                              "retrying in background")
+                self.log.warn("upstream not reachable yet; watching "
+                              "replication in the background")
+                self._transition_task = \
+                    asyncio.get_running_loop().create_task(
+                        self._standby_watch())
         if need_restore:
             await self._restore_and_start(upstream, restore_peer)
+
+    async def _standby_watch(self) -> None:
+        """Background: if a standby that could not reach its upstream turns
+        out to be DIVERGED once the upstream is back, force the FSM to
+        re-issue the transition (which will restore)."""
+        deadline = time.monotonic() + self.replication_timeout_s
+        try:
+            while time.monotonic() < deadline:
+                await asyncio.sleep(0.5)
+                try:
+                    status = await self.engine.status()
+                except Exception:
+                    continue
+                ustat = status.get("upstream_status")
+                if ustat == "streaming":
+                    return
+                if ustat == "diverged":
+                    # the FSM clears its applied config on db errors and
+                    # re-issues the transition, which will restore
+                    self.log.warn("divergence detected by watchdog; "
+                                  "forcing re-transition")
+                    self._emit("error",
+                               RuntimeError("standby diverged from upstream"))
+                    return
+        except asyncio.CancelledError:
+            raise
 
     async def _await_streaming(self, timeout_s: float) -> str:
         deadline = time.monotonic() + timeout_s
@@ -375,27 +406,15 @@ class DbManager:
         await self._stop_db()
         self.restore_client = RestoreClient(self.store, self.ip,
                                             log=self.log)
-        # only data worth keeping is isolated; an empty/uninitialized
-        # dataset is simply replaced.  Restore is retried with backoff —
-        # the restore peer's backup service may still be coming up
-        # (ref restore retry budget, lib/adm.js:71,1613-1617)
+        # ONE attempt per reconfigure, like the reference's _standby
+        # (ref :1375-1413): a failure propagates and the FSM re-issues the
+        # transition on its next evaluation with a FRESH restorePeer —
+        # critical when the restore peer itself just died in a failover.
+        # Only data worth keeping is isolated; an empty/uninitialized
+        # dataset is simply replaced.
         isolate = self.engine.initialized()
-        attempt = 0
-        while True:
-            attempt += 1
-            try:
-                await self.restore_client.restore(
-                    restore_peer["backupUrl"], isolate=isolate)
-                break
-            except asyncio.CancelledError:
-                raise
-            except Exception as exc:
-                if attempt >= 5:
-                    raise
-                self.log.warn("restore attempt failed; retrying",
-                              attempt=attempt, err=exc)
-                isolate = False  # already isolated (or nothing to keep)
-                await asyncio.sleep(min(2.0 ** attempt * 0.25, 5.0))
+        await self.restore_client.restore(restore_peer["backupUrl"],
+                                          isolate=isolate)
         self.engine.post_restore_fixup()
         self.engine.write_conf("standby", upstream_url=upstream["pgUrl"])
         await self._start_db()

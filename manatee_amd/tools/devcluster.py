@@ -321,6 +321,30 @@ class DevCluster:
                 return p
         raise KeyError(peer_id)
 
+    async def rebuild_peer(self, peer: DevPeer,
+                           timeout_s: float = 120.0) -> None:
+        """The ``manatee-adm rebuild`` flow for a dead/deposed peer
+        (ref lib/adm.js:1319-1684): stop it, destroy its dataset (it is
+        deposed — its WAL may have diverged), remove it from the deposed
+        list, restart it; it restores from the primary and rejoins as an
+        async."""
+        import shutil
+        from ..adm import core as adm
+        peer.kill9()
+        shutil.rmtree(peer.store_dir, ignore_errors=True)
+        zk = await adm.create_zk_client(self.zk_conn_str)
+        try:
+            state, version = await adm.get_state(zk, self.shard_path)
+            if state and any(d["id"] == peer.id
+                             for d in state.get("deposed", [])):
+                await adm.reap(zk, self.shard_path, peer_id=peer.id)
+        finally:
+            await zk.close()
+        peer.start()
+        await self.wait_cluster(
+            lambda s: any(a["id"] == peer.id for a in s.get("async", [])),
+            timeout_s=timeout_s, what="rebuilt peer rejoining as async")
+
     async def wait_writable(self, timeout_s: float = 60.0) -> DevPeer:
         """Wait until the cluster primary accepts a write; returns it."""
         deadline = time.monotonic() + timeout_s
