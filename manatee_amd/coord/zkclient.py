@@ -276,6 +276,9 @@ class ZkClient:
         if etype in (jute.EVENT_NODE_CHILDREN_CHANGED,
                      jute.EVENT_NODE_DELETED):
             cbs.extend(self._child_watches.pop(path, []))
+        if not cbs:
+            self.log.warn("watch event with no registered callback",
+                          etype=etype, path=path)
         for cb in cbs:
             try:
                 cb(etype, path)
@@ -351,9 +354,19 @@ class ZkClient:
                        ) -> Tuple[bytes, Stat]:
         def build(w: Writer):
             w.ustring(path).boolean(watch is not None)
-        r = await self._call(jute.OP_GETDATA, build)
+        # register BEFORE sending: the read loop runs as its own task, so a
+        # notification frame arriving right behind the response could be
+        # dispatched before this coroutine resumes (watch would be lost)
         if watch is not None:
             self._data_watches.setdefault(path, []).append(watch)
+        try:
+            r = await self._call(jute.OP_GETDATA, build)
+        except ZkError:
+            if watch is not None:
+                lst = self._data_watches.get(path, [])
+                if watch in lst:
+                    lst.remove(watch)
+            raise
         data = r.buffer() or b""
         return data, Stat.read(r)
 
@@ -369,9 +382,18 @@ class ZkClient:
                            ) -> Tuple[List[str], Stat]:
         def build(w: Writer):
             w.ustring(path).boolean(watch is not None)
-        r = await self._call(jute.OP_GETCHILDREN2, build)
+        # register BEFORE sending (see get_data): avoids losing an event
+        # delivered between the response and this coroutine resuming
         if watch is not None:
             self._child_watches.setdefault(path, []).append(watch)
+        try:
+            r = await self._call(jute.OP_GETCHILDREN2, build)
+        except ZkError:
+            if watch is not None:
+                lst = self._child_watches.get(path, [])
+                if watch in lst:
+                    lst.remove(watch)
+            raise
         n = r.int32()
         children = [r.ustring() or "" for _ in range(max(n, 0))]
         return children, Stat.read(r)
