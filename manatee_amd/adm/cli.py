@@ -1,0 +1,647 @@
+"""manatee-adm — operator CLI (ref bin/manatee-adm).
+
+Subcommand groups mirror the reference's (bin/manatee-adm:94-122):
+
+  status commands:   status, show, peers, pg-status, verify, history
+  state commands:    zk-state (state), zk-active (active), freeze,
+                     unfreeze, set-onwm, reap, state-backfill, check-lock
+  peer commands:     rebuild, promote, clear-promote
+  other:             version, help
+
+Env fallbacks: ``ZK_IPS`` for -z/--zk, ``SHARD`` for -s/--shard,
+``MANATEE_SITTER_CONFIG`` for -c/--config (ref bin/manatee-adm:31-88,
+docs/man/manatee-adm.md:502-514).  ``MANATEE_ADM_TEST_STATE`` loads a
+fixture instead of touching ZK/db — the golden-test seam.
+"""
+
+from __future__ import annotations
+
+import argparse
+import asyncio
+import json
+import os
+import shutil
+import sys
+import time
+from typing import List, Optional
+
+from .. import __version__
+from ..common.logging import null_logger
+from ..db.waldb.client import WaldbClient
+from ..fsm import state as st
+from . import core as adm
+from . import details as det
+
+RESTORE_ATTEMPTS = 5          # ref lib/adm.js:71
+
+
+class UsageError(Exception):
+    pass
+
+
+def _fail(msg: str) -> int:
+    print("manatee-adm: %s" % msg, file=sys.stderr)
+    return 1
+
+
+def _need(ns, attr: str, env: str, what: str) -> str:
+    val = getattr(ns, attr, None) or os.environ.get(env)
+    if not val:
+        raise UsageError("%s required (or set %s)" % (what, env))
+    return val
+
+
+async def _with_zk(ns, fn):
+    zk = await adm.create_zk_client(_need(ns, "zk", "ZK_IPS", "-z/--zk"),
+                                    log=null_logger())
+    try:
+        return await fn(zk)
+    finally:
+        await zk.close()
+
+
+async def _details(ns) -> det.ClusterDetails:
+    fx = det.fixture_path()
+    if fx:
+        return det.load_fixture(fx)
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+
+    async def go(zk):
+        return await det.load_cluster_details(
+            zk, shard, zk_conn=_need(ns, "zk", "ZK_IPS", "-z/--zk"))
+    return await _with_zk(ns, go)
+
+
+def _print_issues(cd: det.ClusterDetails, stream, leading_nl: bool) -> None:
+    """ref printClusterIssues bin/manatee-adm:1487-1500."""
+    if leading_nl and (cd.errors or cd.warnings):
+        print("", file=stream)
+    for e in cd.errors:
+        print("error: %s" % e.split("\n")[0], file=stream)
+    for w in cd.warnings:
+        print("warning: %s" % w.split("\n")[0], file=stream)
+
+
+def _columns(ns, default: List[str]) -> List[str]:
+    if not getattr(ns, "columns", None):
+        return default
+    cols = []
+    for chunk in ns.columns:
+        cols.extend(c.strip() for c in chunk.split(",") if c.strip())
+    for c in cols:
+        if c not in det.COLUMNS:
+            raise UsageError("unknown column %r (have: %s)"
+                             % (c, ", ".join(sorted(det.COLUMNS))))
+    return cols
+
+
+def _check_role(ns) -> Optional[str]:
+    role = getattr(ns, "role", None)
+    if role and role not in det.ROLES:
+        raise UsageError('unsupported value for --role: "%s"' % role)
+    return role
+
+
+# ------------------------------------------------------------ subcommands
+
+def cmd_version(ns) -> int:
+    print(__version__)
+    return 0
+
+
+async def cmd_peers(ns) -> int:
+    cd = await _details(ns)
+    cols = _columns(ns, det.PEERS_COLUMNS)
+    sys.stdout.write(det.render_table(
+        cols, cd.table_rows(cols, role=_check_role(ns)),
+        header=not ns.omit_header))
+    return 0
+
+
+async def cmd_pg_status(ns) -> int:
+    cols = _columns(ns, det.STATUS_COLUMNS_WIDE if ns.wide
+                    else det.STATUS_COLUMNS)
+    role = _check_role(ns)
+    period = ns.period
+    count = ns.count if ns.count is not None else (None if period else 1)
+    shown = 0
+    while True:
+        cd = await _details(ns)
+        sys.stdout.write(det.render_table(
+            cols, cd.table_rows(cols, role=role),
+            header=not ns.omit_header))
+        _print_issues(cd, sys.stderr, leading_nl=True)
+        shown += 1
+        if count is not None and shown >= count:
+            break
+        await asyncio.sleep(period)
+    return 0
+
+
+def _show_header(cd: det.ClusterDetails) -> None:
+    """ref do_show: zookeeper/cluster/generation/mode/freeze lines."""
+    print("zookeeper:   %s" % (cd.zk_conn or "-"))
+    print("cluster:     %s" % cd.shard)
+    print("generation:  %s (%s)" % (cd.generation, cd.init_wal))
+    print("mode:        %s" % ("singleton (one-node-write)"
+                               if cd.singleton else "normal"))
+    if cd.freeze:
+        print("freeze:      frozen since %s" % cd.freeze.get("date"))
+        print("freeze info: %s" % cd.freeze.get("reason"))
+    else:
+        print("freeze:      not frozen")
+    if cd.promote:
+        print("promote:     pending for %s (expires %s)"
+              % (cd.promote.get("id"), cd.promote.get("expireTime")))
+
+
+async def cmd_show(ns) -> int:
+    cd = await _details(ns)
+    _show_header(cd)
+    print()
+    if ns.verbose:
+        cols = det.PEERS_COLUMNS
+        sys.stdout.write(det.render_table(cols, cd.table_rows(cols)))
+        print()
+    cols = det.STATUS_COLUMNS
+    sys.stdout.write(det.render_table(cols, cd.table_rows(cols)))
+    _print_issues(cd, sys.stdout, leading_nl=True)
+    return 0
+
+
+async def cmd_verify(ns) -> int:
+    try:
+        cd = await _details(ns)
+    except Exception:
+        print("error: failed to fetch cluster state")
+        return 1
+    _print_issues(cd, sys.stdout, leading_nl=False)
+    if cd.errors or cd.warnings:
+        return 1
+    if ns.verbose:
+        print("all checks passed")
+    return 0
+
+
+async def cmd_status(ns) -> int:
+    """JSON summary (ref status lib/adm.js:997-1027): every shard when -s
+    is not given, keyed by shard name."""
+    fx = det.fixture_path()
+    if fx:
+        cd = det.load_fixture(fx)
+        print(json.dumps({cd.shard: _status_obj(cd)}, indent=2,
+                         sort_keys=True))
+        return 0
+
+    async def go(zk):
+        shard = getattr(ns, "shard", None) or os.environ.get("SHARD")
+        shards = [shard] if shard else await adm.get_shards(zk)
+        out = {}
+        for sh in shards:
+            try:
+                cd = await det.load_cluster_details(
+                    zk, sh, zk_conn=ns.zk or os.environ.get("ZK_IPS", ""))
+                out[sh] = _status_obj(cd)
+            except adm.AdmError as exc:
+                out[sh] = {"error": str(exc)}
+        print(json.dumps(out, indent=2, sort_keys=True))
+        return 0
+    return await _with_zk(ns, go)
+
+
+def _status_obj(cd: det.ClusterDetails) -> dict:
+    def peer_obj(pid: str) -> dict:
+        pd = cd.peers[pid]
+        o = dict(pd.ident)
+        o["online"] = pd.online
+        if pd.db_error:
+            o["error"] = pd.db_error
+        row = pd.first_repl()
+        if row:
+            o["repl"] = row
+        if pd.lag_s is not None:
+            o["lag_s"] = round(pd.lag_s, 3)
+        return o
+    out = {
+        "generation": cd.generation,
+        "initWal": cd.init_wal,
+        "oneNodeWriteMode": cd.singleton,
+        "freeze": cd.freeze,
+        "primary": peer_obj(cd.primary_id) if cd.primary_id else None,
+        "sync": peer_obj(cd.sync_id) if cd.sync_id else None,
+        "async": [peer_obj(a) for a in cd.async_ids],
+        "deposed": [peer_obj(d) for d in cd.deposed_ids],
+        "errors": cd.errors,
+        "warnings": cd.warnings,
+    }
+    return out
+
+
+async def cmd_zk_state(ns) -> int:
+    fx = det.fixture_path()
+    if fx:
+        cd = det.load_fixture(fx)
+        print(json.dumps(cd.state, indent=2, sort_keys=True))
+        return 0
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+
+    async def go(zk):
+        state, version = await adm.get_state(zk, shard)
+        if state is None:
+            return _fail("no cluster state for shard %r" % shard)
+        state["_zkVersion"] = version
+        print(json.dumps(state, indent=2, sort_keys=True))
+        return 0
+    return await _with_zk(ns, go)
+
+
+async def cmd_zk_active(ns) -> int:
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+
+    async def go(zk):
+        actives = await adm.get_active(zk, shard)
+        print(json.dumps(actives, indent=2, sort_keys=True))
+        return 0
+    return await _with_zk(ns, go)
+
+
+async def cmd_history(ns) -> int:
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+
+    async def go(zk):
+        entries = await adm.get_history(zk, shard)
+        annotated = adm.annotate_history(entries)
+        if ns.json:
+            for e in annotated:
+                print(json.dumps(e, sort_keys=True))
+            return 0
+        for e in annotated:
+            s = e["state"]
+            when = st.iso8601(e["time"] / 1000.0) if e["time"] else "-"
+            line = "%-4s %-24s gen %-3s primary %-8s sync %-8s" % (
+                e["zkSeq"], when, s.get("generation"),
+                (s.get("primary") or {}).get("id", "-")[:8],
+                ((s.get("sync") or {}).get("id", "-") or "-")[:8])
+            asyncs = ",".join(a["id"][:8] for a in s.get("async") or [])
+            deposed = ",".join(d["id"][:8] for d in s.get("deposed") or [])
+            line += " async [%s]" % asyncs
+            if deposed:
+                line += " deposed [%s]" % deposed
+            if s.get("freeze"):
+                line += " FROZEN"
+            print(line)
+            for note in e.get("notes", []):
+                print("         %s" % note)
+            for viol in e.get("violations", []):
+                print("         VIOLATION: %s" % viol)
+        return 0
+    return await _with_zk(ns, go)
+
+
+async def cmd_freeze(ns) -> int:
+    if not ns.reason:
+        raise UsageError("freeze requires a reason (-r)")
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+
+    async def go(zk):
+        await adm.freeze(zk, shard, ns.reason)
+        print("Frozen.")
+        return 0
+    return await _with_zk(ns, go)
+
+
+async def cmd_unfreeze(ns) -> int:
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+
+    async def go(zk):
+        try:
+            await adm.unfreeze(zk, shard)
+        except adm.AdmError as exc:
+            return _fail(str(exc))
+        print("Unfrozen.")
+        return 0
+    return await _with_zk(ns, go)
+
+
+async def cmd_reap(ns) -> int:
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+    if not ns.id and not ns.zonename:
+        raise UsageError("reap requires -i/--id or -n/--zonename")
+
+    async def go(zk):
+        try:
+            await adm.reap(zk, shard, peer_id=ns.id, zonename=ns.zonename)
+        except adm.AdmError as exc:
+            return _fail(str(exc))
+        print("Reaped.")
+        return 0
+    return await _with_zk(ns, go)
+
+
+async def cmd_set_onwm(ns) -> int:
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+    if ns.mode not in ("on", "off"):
+        raise UsageError("set-onwm requires -m on|off")
+    if not ns.yes:
+        raise UsageError("set-onwm is dangerous; confirm with -y")
+
+    async def go(zk):
+        try:
+            await adm.set_onwm(zk, shard, ns.mode)
+        except adm.AdmError as exc:
+            return _fail(str(exc))
+        print("one-node-write mode: %s" % ns.mode)
+        return 0
+    return await _with_zk(ns, go)
+
+
+async def cmd_state_backfill(ns) -> int:
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+    if not ns.yes:
+        raise UsageError("state-backfill rewrites cluster state; confirm "
+                         "with -y")
+
+    async def go(zk):
+        try:
+            state = await adm.state_backfill(zk, shard)
+        except adm.AdmError as exc:
+            return _fail(str(exc))
+        print(json.dumps(state, indent=2, sort_keys=True))
+        return 0
+    return await _with_zk(ns, go)
+
+
+async def cmd_check_lock(ns) -> int:
+    """Exit 1 (with the lock's contents) if the named ZK path exists
+    (ref checkLock lib/adm.js:2049-2061)."""
+    if not ns.path:
+        raise UsageError("check-lock requires -p/--path")
+
+    async def go(zk):
+        from ..coord import jute
+        try:
+            data, _ = await zk.get_data(ns.path)
+        except jute.ZkError as exc:
+            if exc.code == jute.ZNONODE:
+                return 0
+            raise
+        print(data.decode("utf-8", "replace"))
+        return 1
+    return await _with_zk(ns, go)
+
+
+async def cmd_promote(ns) -> int:
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+    role = ns.role or "sync"
+
+    async def go(zk):
+        try:
+            state = await adm.request_promote(
+                zk, shard, role=role, peer_id=ns.id, zonename=ns.zonename,
+                async_index=ns.async_index)
+        except adm.AdmError as exc:
+            return _fail(str(exc))
+        promote = state["promote"]
+        print("promote requested: %s (%s), expires %s"
+              % (promote["id"], promote["role"], promote["expireTime"]))
+        # watch until consumed or expired (ref promote :1693-2014)
+        deadline = time.monotonic() + 60
+        while time.monotonic() < deadline:
+            cur, _ = await adm.get_state(zk, shard)
+            if cur is None:
+                break
+            if "promote" not in cur:
+                if cur["generation"] != state["generation"]:
+                    print("promotion complete (generation %s)"
+                          % cur["generation"])
+                else:
+                    print("promote request consumed")
+                return 0
+            exp = cur["promote"].get("expireTime", "")
+            if exp and exp < st.iso8601():
+                print("promote request expired without being applied",
+                      file=sys.stderr)
+                return 1
+            await asyncio.sleep(0.5)
+        print("gave up waiting for the promotion", file=sys.stderr)
+        return 1
+    return await _with_zk(ns, go)
+
+
+async def cmd_clear_promote(ns) -> int:
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
+
+    async def go(zk):
+        try:
+            await adm.clear_promote(zk, shard)
+        except adm.AdmError as exc:
+            return _fail(str(exc))
+        print("Cleared.")
+        return 0
+    return await _with_zk(ns, go)
+
+
+async def cmd_rebuild(ns) -> int:
+    """Operator rebuild of THIS peer (ref rebuild lib/adm.js:1319-1684):
+    refuse on the primary; require the peer's sitter to be stopped (its
+    election ephemeral gone); isolate (or destroy, if deposed) the data
+    store; remove the peer from the deposed list; restart the sitter
+    (via --start-cmd) and watch the restore until the peer rejoins."""
+    cfg_path = ns.config or os.environ.get("MANATEE_SITTER_CONFIG")
+    if not cfg_path:
+        raise UsageError("rebuild requires -c/--config (or "
+                         "MANATEE_SITTER_CONFIG)")
+    with open(cfg_path) as f:
+        cfg = json.load(f)
+    peer_id = "%s:%s:%s" % (cfg["ip"], cfg["postgresPort"],
+                            cfg["backupPort"])
+    shard = cfg["shardPath"]
+    ns.zk = ns.zk or cfg["zkCfg"]["connStr"]
+    session_ms = cfg["zkCfg"].get("opts", {}).get("sessionTimeout", 60000)
+    store_cfg = cfg["postgresMgrCfg"]["storageCfg"]
+
+    async def go(zk):
+        state, _ = await adm.get_state(zk, shard)
+        if state is None:
+            return _fail("no cluster state for shard %r" % shard)
+        if state.get("primary", {}).get("id") == peer_id:
+            return _fail("refusing to rebuild the PRIMARY peer; promote "
+                         "another peer first")
+        deposed = any(d.get("id") == peer_id
+                      for d in state.get("deposed") or [])
+        if not ns.yes:
+            return _fail("rebuild destroys this peer's local data%s; "
+                         "confirm with -y"
+                         % (" (peer is DEPOSED)" if deposed else ""))
+        if ns.stop_cmd:
+            proc = await asyncio.create_subprocess_shell(ns.stop_cmd)
+            await proc.wait()
+        # wait for the peer's ephemeral election node to expire
+        # (ref :1433-1478 — up to 1.5x the session timeout)
+        deadline = time.monotonic() + 1.5 * session_ms / 1000.0
+        while True:
+            actives = await adm.get_active(zk, shard)
+            if not any(a["id"] == peer_id for a in actives):
+                break
+            if time.monotonic() > deadline:
+                return _fail("peer %s still has a live ZK session; stop "
+                             "its sitter first (or pass --stop-cmd)"
+                             % peer_id)
+            await asyncio.sleep(0.5)
+        # isolate or destroy the local store
+        from ..storage import make_store
+        store = make_store(store_cfg, log=null_logger())
+        if await store.exists():
+            if deposed:
+                await store.destroy()
+                print("destroyed local store (peer was deposed)")
+            else:
+                moved = await store.isolate("autorebuild")
+                print("isolated local store%s"
+                      % (" to %s" % moved if moved else ""))
+        if deposed:
+            try:
+                await adm.reap(zk, shard, peer_id=peer_id)
+                print("removed %s from the deposed list" % peer_id)
+            except adm.AdmError:
+                pass
+        if not ns.start_cmd:
+            print("local state cleared; restart the sitter to restore "
+                  "from the primary")
+            return 0
+        proc = await asyncio.create_subprocess_shell(ns.start_cmd)
+        await proc.wait()
+        # watch the peer's status server until it is back as sync/async
+        # (ref :1550-1678; restore retry budget RESTORE_ATTEMPTS)
+        status_url = "http://%s:%d" % (cfg["ip"], cfg["postgresPort"] + 1)
+        print("waiting for %s to restore and rejoin (watch %s/restore)"
+              % (peer_id, status_url))
+        deadline = time.monotonic() + ns.timeout
+        while time.monotonic() < deadline:
+            cur, _ = await adm.get_state(zk, shard)
+            if cur is not None:
+                ids = [a.get("id") for a in cur.get("async") or []]
+                if cur.get("sync"):
+                    ids.append(cur["sync"].get("id"))
+                if peer_id in ids:
+                    print("peer %s rejoined the cluster" % peer_id)
+                    return 0
+            await asyncio.sleep(1.0)
+        return _fail("peer did not rejoin within %ds" % ns.timeout)
+    return await _with_zk(ns, go)
+
+
+# --------------------------------------------------------------- plumbing
+
+def _mk_parser() -> argparse.ArgumentParser:
+    p = argparse.ArgumentParser(
+        prog="manatee-adm",
+        description="Administer a manatee shard (clean-room rebuild; "
+                    "ref bin/manatee-adm)")
+    sub = p.add_subparsers(dest="cmd")
+
+    def add(name: str, fn, aliases=(), **kw):
+        sp = sub.add_parser(name, aliases=list(aliases), **kw)
+        sp.set_defaults(fn=fn)
+        sp.add_argument("-z", "--zk", help="ZooKeeper connection string "
+                        "(env ZK_IPS)")
+        sp.add_argument("-s", "--shard", help="shard name (env SHARD)")
+        return sp
+
+    add("version", cmd_version, help="print the version")
+    add("status", cmd_status, help="JSON summary of every shard")
+
+    sp = add("peers", cmd_peers, help="table of peers")
+    sp.add_argument("-H", "--omitHeader", dest="omit_header",
+                    action="store_true")
+    sp.add_argument("-o", "--columns", action="append")
+    sp.add_argument("-r", "--role")
+
+    sp = add("pg-status", cmd_pg_status, aliases=["db-status"],
+             help="table of database status per peer")
+    sp.add_argument("-H", "--omitHeader", dest="omit_header",
+                    action="store_true")
+    sp.add_argument("-o", "--columns", action="append")
+    sp.add_argument("-r", "--role")
+    sp.add_argument("-w", "--wide", action="store_true")
+    sp.add_argument("period", nargs="?", type=int, default=None)
+    sp.add_argument("count", nargs="?", type=int, default=None)
+
+    sp = add("show", cmd_show, help="summary + status table")
+    sp.add_argument("-v", "--verbose", action="store_true")
+
+    sp = add("verify", cmd_verify, help="health checks; exit 1 on issues")
+    sp.add_argument("-v", "--verbose", action="store_true")
+
+    add("zk-state", cmd_zk_state, aliases=["state"],
+        help="raw cluster state JSON")
+    add("zk-active", cmd_zk_active, aliases=["active"],
+        help="live election members")
+
+    sp = add("history", cmd_history,
+             help="cluster state history with transition checks")
+    sp.add_argument("-j", "--json", action="store_true")
+
+    sp = add("freeze", cmd_freeze, help="freeze cluster transitions")
+    sp.add_argument("-r", "--reason")
+    add("unfreeze", cmd_unfreeze, help="resume cluster transitions")
+
+    sp = add("reap", cmd_reap, help="drop a peer from the deposed list")
+    sp.add_argument("-i", "--id")
+    sp.add_argument("-n", "--zonename")
+
+    sp = add("set-onwm", cmd_set_onwm, help="toggle one-node-write mode")
+    sp.add_argument("-m", "--mode", choices=("on", "off"))
+    sp.add_argument("-y", "--yes", action="store_true")
+
+    sp = add("state-backfill", cmd_state_backfill,
+             help="synthesize v2 state from the election order")
+    sp.add_argument("-y", "--yes", action="store_true")
+
+    sp = add("check-lock", cmd_check_lock,
+             help="exit 1 if the named ZK lock path exists")
+    sp.add_argument("-p", "--path")
+
+    sp = add("promote", cmd_promote, help="request a peer promotion")
+    sp.add_argument("-i", "--id")
+    sp.add_argument("-n", "--zonename")
+    sp.add_argument("--role", choices=("sync", "async"))
+    sp.add_argument("--asyncIndex", dest="async_index", type=int)
+
+    add("clear-promote", cmd_clear_promote,
+        help="remove a pending promote request")
+
+    sp = add("rebuild", cmd_rebuild, help="rebuild THIS peer from the "
+             "primary's backup")
+    sp.add_argument("-c", "--config",
+                    help="sitter config (env MANATEE_SITTER_CONFIG)")
+    sp.add_argument("-y", "--yes", action="store_true")
+    sp.add_argument("--stop-cmd", help="command to stop the local sitter")
+    sp.add_argument("--start-cmd",
+                    help="command to start the local sitter")
+    sp.add_argument("--timeout", type=int, default=300)
+    return p
+
+
+def main(argv=None) -> int:
+    parser = _mk_parser()
+    ns = parser.parse_args(argv)
+    if not getattr(ns, "fn", None):
+        parser.print_help()
+        return 2
+    try:
+        res = ns.fn(ns)
+        if asyncio.iscoroutine(res):
+            res = asyncio.run(res)
+        return int(res or 0)
+    except UsageError as exc:
+        print("manatee-adm: %s" % exc, file=sys.stderr)
+        return 2
+    except KeyboardInterrupt:
+        return 130
+    except BrokenPipeError:
+        return 0
+
+
+if __name__ == "__main__":
+    sys.exit(main())

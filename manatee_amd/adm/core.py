@@ -143,6 +143,93 @@ async def get_history(zk: ZkClient, shard: str) -> List[dict]:
     return out
 
 
+def annotate_history(entries: List[dict]) -> List[dict]:
+    """Attach human annotations + legal-transition checks to each history
+    entry (ref annotateHistoryNode lib/adm.js:2296-2416 — that function
+    encodes the FSM's safety rules: the generation never goes backwards,
+    a new primary must be the previous sync, and a sync change requires a
+    generation bump).  Each returned entry gains ``notes`` and
+    ``violations`` lists."""
+    def pid(ident) -> Optional[str]:
+        return ident.get("id") if ident else None
+
+    def abbr(ident) -> str:
+        return (pid(ident) or "-")[:8]
+
+    out = []
+    last: Optional[dict] = None
+    for e in entries:
+        notes: List[str] = []
+        violations: List[str] = []
+        nst = e["state"]
+        if last is None:
+            notes.append("cluster setup for %s mode"
+                         % ("singleton (one-node-write)"
+                            if nst.get("oneNodeWriteMode")
+                            else "normal (multi-peer)"))
+            out.append({**e, "notes": notes, "violations": violations})
+            last = nst
+            continue
+        lst = last
+        ngen, lgen = nst.get("generation"), lst.get("generation")
+        if ngen < lgen:
+            violations.append("gen number went backwards")
+        elif not lst.get("oneNodeWriteMode") and nst.get("oneNodeWriteMode"):
+            violations.append("unsupported transition from multi-peer mode "
+                              "to singleton (one-node-write) mode")
+        elif lst.get("oneNodeWriteMode") and not nst.get("oneNodeWriteMode"):
+            notes.append("cluster transitioned from singleton "
+                         "(one-node-write) mode to multi-peer mode")
+        elif pid(nst.get("primary")) != pid(lst.get("primary")):
+            if ngen == lgen:
+                violations.append("new primary, but same gen number")
+            elif (lst.get("sync") is None
+                  or pid(nst["primary"]) != pid(lst["sync"])):
+                violations.append("new primary was not previous sync")
+            else:
+                notes.append("sync (%s) took over as primary (from %s)"
+                             % (abbr(nst["primary"]), abbr(lst["primary"])))
+        elif ngen > lgen:
+            if lst.get("sync") is None and not lst.get("oneNodeWriteMode"):
+                notes.append('sync "%s" added' % abbr(nst.get("sync")))
+            elif pid(nst.get("sync")) == pid(lst.get("sync")):
+                violations.append("gen number changed, but primary and "
+                                  "sync did not")
+            else:
+                notes.append("primary (%s) selected new sync (was %s, "
+                             "now %s)" % (abbr(nst["primary"]),
+                                          abbr(lst.get("sync")),
+                                          abbr(nst.get("sync"))))
+        elif pid(nst.get("sync")) != pid(lst.get("sync")):
+            violations.append("sync changed, but gen number did not")
+        else:
+            if nst.get("freeze") and not lst.get("freeze"):
+                notes.append("cluster frozen: %s"
+                             % nst["freeze"].get("reason"))
+            elif lst.get("freeze") and not nst.get("freeze"):
+                notes.append("cluster unfrozen")
+            news = {pid(a) for a in nst.get("async") or []}
+            olds = {pid(a) for a in lst.get("async") or []}
+            for a in sorted(news - olds):
+                notes.append('async "%s" added' % a[:8])
+            for a in sorted(olds - news):
+                notes.append('async "%s" removed' % a[:8])
+            newd = {pid(d) for d in nst.get("deposed") or []}
+            oldd = {pid(d) for d in lst.get("deposed") or []}
+            for d in sorted(newd - oldd):
+                notes.append('"%s" deposed' % d[:8])
+            for d in sorted(oldd - newd):
+                notes.append('"%s" no longer deposed' % d[:8])
+            if nst.get("promote") and not lst.get("promote"):
+                notes.append('promote requested for "%s"'
+                             % (nst["promote"].get("id") or "-")[:8])
+            elif lst.get("promote") and not nst.get("promote"):
+                notes.append("promote request cleared")
+        out.append({**e, "notes": notes, "violations": violations})
+        last = nst
+    return out
+
+
 # ----------------------------------------------------------------- operator ops
 
 async def freeze(zk: ZkClient, shard: str, reason: str) -> dict:
