@@ -20,7 +20,6 @@ import argparse
 import asyncio
 import json
 import os
-import shutil
 import sys
 import time
 from typing import List, Optional
@@ -489,8 +488,8 @@ async def cmd_rebuild(ns) -> int:
                              % peer_id)
             await asyncio.sleep(0.5)
         # isolate or destroy the local store
-        from ..storage import make_store
-        store = make_store(store_cfg, log=null_logger())
+        from ..storage import open_store
+        store = open_store(store_cfg, log=null_logger())
         if await store.exists():
             if deposed:
                 await store.destroy()

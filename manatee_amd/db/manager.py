@@ -146,6 +146,15 @@ class DbManager:
             *argv, stdout=asyncio.subprocess.DEVNULL,
             stderr=asyncio.subprocess.DEVNULL,
             start_new_session=True)
+        # record the child pid IMMEDIATELY: the db writes its own pid file
+        # only once it is up, and anything that needs to SIGKILL the whole
+        # peer (tests, operators) must not race that window
+        try:
+            with open(os.path.join(self.engine.data_dir, "db_child.pid"),
+                      "w") as f:
+                f.write(str(self._proc.pid))
+        except (OSError, AttributeError):
+            pass
         self._proc_monitor = asyncio.get_running_loop().create_task(
             self._monitor_proc(self._proc))
         # poll until the db answers (ref _start 1 Hz poll :1760-1794;

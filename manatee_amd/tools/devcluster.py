@@ -125,19 +125,28 @@ class DevPeer:
         self.sitter_proc = self._spawn("manatee_amd.daemons.sitter",
                                        "sitter.json", "sitter.log")
 
+    def db_pids(self) -> List[int]:
+        """Candidate db pids: the manager-written db_child.pid (written at
+        spawn time, so it can never lag the child) plus the db's own pid
+        file (covers children of sitters from earlier incarnations)."""
+        data = os.path.join(self.store_dir, "live", "data")
+        pids = []
+        for name in ("db_child.pid", "waldb.pid"):
+            try:
+                with open(os.path.join(data, name)) as f:
+                    pids.append(int(f.read().split()[0]))
+            except (OSError, ValueError, IndexError):
+                pass
+        return pids
+
     def db_pid(self) -> Optional[int]:
-        pid_file = os.path.join(self.store_dir, "live", "data", "waldb.pid")
-        try:
-            with open(pid_file) as f:
-                return int(f.read().split()[0])
-        except (OSError, ValueError, IndexError):
-            return None
+        pids = self.db_pids()
+        return pids[0] if pids else None
 
     def kill9(self) -> None:
         """SIGKILL the whole peer: sitter process group + db child +
         backupserver — the integ-test failure mode
         (ref test/integ.test.js primaryDeath et al)."""
-        db_pid = self.db_pid()
         for proc in (self.sitter_proc, self.backup_proc):
             if proc is not None and proc.poll() is None:
                 try:
@@ -145,7 +154,10 @@ class DevPeer:
                 except ProcessLookupError:
                     pass
                 proc.wait()
-        if db_pid is not None:
+        # read the db pids only AFTER the sitter is dead: the sitter can
+        # respawn the db at any moment, so a pid captured earlier can go
+        # stale and the fresh child would survive holding the port
+        for db_pid in self.db_pids():
             try:
                 os.killpg(db_pid, signal.SIGKILL)
             except (ProcessLookupError, PermissionError):
