@@ -147,7 +147,7 @@ class WaldbEngine(Engine):
                           waldb_server.PROMOTE_TRIGGER), "w").close()
 
     def post_restore_fixup(self) -> None:
-        for name in ("waldb.pid",):
+        for name in ("waldb.pid", "db_child.pid"):
             try:
                 os.unlink(os.path.join(self.data_dir, name))
             except FileNotFoundError:

@@ -256,13 +256,19 @@ class DbManager:
             self.engine.write_promote_trigger()
         await self._restart_db()
         self.writable = bool(onwm)
-        try:
-            await self.store.snapshot()   # for future bootstraps (ref :1158)
-        except Exception as exc:
-            self.log.warn("post-transition snapshot failed", err=exc)
+        # snapshot for future bootstraps (ref :1158) — in the background:
+        # the reference's zfs snapshot is O(1), ours copies data, and the
+        # failover-to-writable path must not wait on it
+        asyncio.get_running_loop().create_task(self._background_snapshot())
         if not onwm:
             self._transition_task = asyncio.get_running_loop().create_task(
                 self._wait_for_standby(downstream))
+
+    async def _background_snapshot(self) -> None:
+        try:
+            await self.store.snapshot()
+        except Exception as exc:
+            self.log.warn("post-transition snapshot failed", err=exc)
 
     async def _update_standby(self, cfg: dict) -> None:
         """Downstream swap on a running primary: conf + SIGHUP only

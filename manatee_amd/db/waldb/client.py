@@ -99,6 +99,14 @@ class WaldbClient:
             raise WaldbError(resp.get("error", "get failed"))
         return resp.get("v") if resp.get("found") else None
 
+    async def count(self, prefix: Optional[str] = None,
+                    timeout_s: Optional[float] = None) -> int:
+        resp = await self.query({"q": "count", "prefix": prefix},
+                                timeout_s=timeout_s)
+        if not resp.get("ok"):
+            raise WaldbError(resp.get("error", "count failed"))
+        return int(resp["count"])
+
     async def status(self) -> dict:
         resp = await self.query({"q": "status"})
         if not resp.get("ok"):

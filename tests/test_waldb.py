@@ -40,7 +40,7 @@ class Node:
         init_data_dir(self.data_dir)
 
     def write_conf(self, role="primary", upstream=None, sync_name=None,
-                   read_only=False):
+                   read_only=False, extra=None):
         conf = {
             "role": role,
             "listen_ip": "127.0.0.1",
@@ -52,6 +52,7 @@ class Node:
             conf["primary_conninfo"] = "'%s'" % upstream
         if sync_name:
             conf["synchronous_standby_names"] = "'%s'" % sync_name
+        conf.update(extra or {})
         confparser.write(os.path.join(self.data_dir, "waldb.conf"), conf)
 
     def start(self):
@@ -125,8 +126,11 @@ def test_single_node_put_get_and_crash_recovery(tmp_path):
             await c.close()
         run(phase1())
         n.kill9()
-        # corrupt tail: simulate torn write from the dirty kill
-        with open(os.path.join(n.data_dir, "wal.log"), "ab") as f:
+        # corrupt tail: simulate torn write from the dirty kill (the last
+        # WAL segment is the only one that can be torn)
+        segs = sorted(f for f in os.listdir(n.data_dir)
+                      if f.startswith("wal-") and f.endswith(".seg"))
+        with open(os.path.join(n.data_dir, segs[-1]), "ab") as f:
             f.write(b"\x00\x00\x00\x10partial")
         n.start()
 
@@ -353,3 +357,165 @@ def test_acked_writes_survive_primary_kill9(tmp_path):
     finally:
         prim.stop()
         sync.stop()
+
+
+# ----------------------------------------------------- segmented WAL + ckpt
+
+def test_wal_class_segmentation_and_recycling(tmp_path):
+    """Direct Wal unit test: rolling, cross-segment reads, drop_below,
+    truncate_to, torn-tail recovery, WalGone."""
+    from manatee_amd.db.waldb.wal import Wal, WalGone
+
+    d = str(tmp_path / "wal")
+    w = Wal(d, segment_bytes=256)
+    w.open()
+    lsns = []
+    payloads = []
+    for i in range(40):
+        p = (b"rec-%03d-" % i) * 4
+        payloads.append(p)
+        lsns.append(w.append(p))
+    w.fsync()
+    assert len(w._segs) > 3, "expected several segments"
+
+    # cross-segment read reassembles the stream
+    whole = w.read(0, 1 << 20)
+    assert len(whole) == w.end
+
+    # reopen + full replay
+    w.close()
+    w2 = Wal(d, segment_bytes=256)
+    seen = []
+    w2.open(replay=lambda lsn, p: seen.append((lsn, p)))
+    assert [p for _, p in seen] == payloads
+    assert w2.end == lsns[-1]
+
+    # checkpoint-style recycling drops whole segments
+    mid = lsns[len(lsns) // 2]
+    w2.drop_below(mid)
+    assert 0 < w2.start <= mid
+    with pytest.raises(WalGone):
+        w2.read(0)
+    # replay_from skips dropped history
+    w2.close()
+    w3 = Wal(d, segment_bytes=256)
+    seen3 = []
+    w3.open(replay=lambda lsn, p: seen3.append(lsn), replay_from=mid)
+    assert all(lsn > mid for lsn in seen3)
+    assert w3.end == lsns[-1]
+
+    # timeline fencing truncation
+    cut = lsns[-5]
+    w3.truncate_to(cut)
+    assert w3.end == cut
+    w3.append(b"after-truncate")
+    assert w3.end > cut
+
+    # torn tail in the LAST segment only is repaired
+    w3.fsync()
+    last_seg = sorted(w3._segs)[-1]
+    with open(os.path.join(d, "wal-%016x.seg" % last_seg), "ab") as f:
+        f.write(b"\x00\x00\x00\x20torn")
+    w3.close()
+    w4 = Wal(d, segment_bytes=256)
+    end = w4.open(replay_from=w4.start)
+    assert end == cut + 8 + len(b"after-truncate")
+    w4.close()
+
+
+def test_checkpoint_bounds_recovery_and_wal_size(tmp_path):
+    """A primary under write load checkpoints, recycles old WAL segments
+    and recovers from checkpoint + tail after kill -9 with all data."""
+    small = {"checkpoint_wal_bytes": "4096", "wal_keep_bytes": "4096",
+             "wal_segment_bytes": "2048"}
+    n = Node(tmp_path, "ckpt")
+    n.init()
+    n.write_conf(role="primary", extra=small)
+    n.start()
+    try:
+        async def fill():
+            c = n.client()
+            for i in range(400):
+                await c.put("k%d" % i, "v" * 50)
+
+            async def ckpted():
+                st = await c.status()
+                return st["checkpoint_lsn"] != "0/00000000" and \
+                    st["wal_start_lsn"] != "0/00000000"
+            await wait_async(ckpted, timeout=15,
+                             what="checkpoint + WAL recycling")
+            st = await c.status()
+            assert st["wal_retained_bytes"] < 64 * 1024
+            assert await c.count(prefix="k") == 400
+            await c.close()
+        run(fill())
+        segs = [f for f in os.listdir(n.data_dir)
+                if f.startswith("wal-")]
+        assert len(segs) < 20, "old segments were not recycled"
+        assert os.path.exists(os.path.join(n.data_dir, "checkpoint.json"))
+
+        n.kill9()
+        n.start()
+
+        async def verify():
+            c = n.client()
+            assert await c.count(prefix="k") == 400
+            for i in (0, 199, 399):
+                assert await c.get("k%d" % i) == "v" * 50
+            await c.close()
+        run(verify())
+    finally:
+        n.stop()
+
+
+def test_follower_behind_recycled_wal_is_refused(tmp_path):
+    """A standby whose position predates the primary's oldest retained
+    segment must be refused with wal-gone and flag itself diverged (the
+    manager then restores it from a snapshot — integ-tested elsewhere)."""
+    small = {"checkpoint_wal_bytes": "2048", "wal_keep_bytes": "1024",
+             "wal_segment_bytes": "1024"}
+    prim = Node(tmp_path, "prim")
+    prim.init()
+    prim.write_conf(role="primary", extra=small)
+    prim.start()
+    try:
+        async def fill():
+            c = prim.client()
+            for i in range(300):
+                await c.put("w%d" % i, "x" * 40)
+
+            async def recycled():
+                st = await c.status()
+                return st["wal_start_lsn"] != "0/00000000"
+            await wait_async(recycled, timeout=15, what="WAL recycling")
+            await c.close()
+        run(fill())
+
+        # fresh standby with an empty WAL (start position 0): the data it
+        # needs is gone — streaming must be refused
+        stand = Node(tmp_path, "stand")
+        # copy the ident so only the WAL position is at issue
+        os.makedirs(stand.data_dir, exist_ok=True)
+        with open(os.path.join(prim.data_dir, "waldb_ident.json")) as f:
+            ident = f.read()
+        with open(os.path.join(stand.data_dir, "waldb_ident.json"),
+                  "w") as f:
+            f.write(ident)
+        stand.write_conf(role="standby",
+                         upstream="127.0.0.1:%d" % prim.port)
+        stand.start()
+        try:
+            async def check():
+                c = stand.client()
+
+                async def diverged():
+                    st = await c.status()
+                    return st["upstream_status"] == "diverged"
+                await wait_async(diverged, timeout=15,
+                                 what="wal-gone -> diverged")
+                await c.close()
+            run(check())
+        finally:
+            stand.stop()
+    finally:
+        prim.stop()
