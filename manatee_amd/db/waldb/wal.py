@@ -28,6 +28,8 @@ import struct
 import zlib
 from typing import Callable, Iterator, List, Optional, Tuple
 
+from ...native import codec as _native
+
 _HDR = struct.Struct(">II")
 _SEG_RE = re.compile(r"^wal-([0-9a-f]{16})\.seg$")
 
@@ -96,22 +98,14 @@ class Wal:
                 continue
             fd = os.open(path, os.O_RDWR)
             try:
-                pos = seg_start
-                while pos + _HDR.size <= seg_start + size:
-                    off = pos - seg_start
-                    hdr = os.pread(fd, _HDR.size, off)
-                    if len(hdr) < _HDR.size:
-                        break
-                    length, crc = _HDR.unpack(hdr)
-                    if length > 64 * 1024 * 1024 or \
-                            pos + _HDR.size + length > seg_start + size:
-                        break
-                    payload = os.pread(fd, length, off + _HDR.size)
-                    if len(payload) != length or zlib.crc32(payload) != crc:
-                        break
-                    pos += _HDR.size + length
-                    if replay is not None and pos > replay_from:
-                        replay(pos, payload)
+                buf = os.pread(fd, size, 0)
+                valid, records = _scan(buf)
+                pos = seg_start + valid
+                if replay is not None:
+                    for off, length in records:
+                        commit = seg_start + off + length
+                        if commit > replay_from:
+                            replay(commit, buf[off:off + length])
                 if pos != seg_start + size:
                     if is_last:
                         os.ftruncate(fd, pos - seg_start)  # torn tail
@@ -151,7 +145,10 @@ class Wal:
     def append(self, payload: bytes) -> int:
         """Append one record; returns its commit LSN."""
         self._maybe_roll()
-        frame = _HDR.pack(len(payload), zlib.crc32(payload)) + payload
+        if _native is not None:
+            frame = _native.encode_frame(payload)
+        else:
+            frame = _HDR.pack(len(payload), zlib.crc32(payload)) + payload
         os.pwrite(self._fd, frame, self.end - self._active_start)
         self.end += len(frame)
         return self.end
@@ -262,9 +259,37 @@ class WalGone(Exception):
         self.have = have
 
 
+def _scan(buf: bytes) -> Tuple[int, List[Tuple[int, int]]]:
+    """Validate a record-stream prefix; returns (valid_bytes,
+    [(payload_offset, payload_len), ...]).  Native-accelerated."""
+    if _native is not None:
+        valid, _count, offsets = _native.scan_records(buf, True)
+        return valid, offsets
+    records: List[Tuple[int, int]] = []
+    pos = 0
+    size = len(buf)
+    while pos + _HDR.size <= size:
+        length, crc = _HDR.unpack_from(buf, pos)
+        if length > 64 * 1024 * 1024 or pos + _HDR.size + length > size:
+            break
+        payload = buf[pos + _HDR.size:pos + _HDR.size + length]
+        if zlib.crc32(payload) != crc:
+            break
+        records.append((pos + _HDR.size, length))
+        pos += _HDR.size + length
+    return pos, records
+
+
 def parse_frames(data: bytes) -> Iterator[Tuple[int, bytes]]:
     """Split a raw replicated chunk into (frame_len, payload) records.
-    The chunk always contains whole frames (senders send record-aligned)."""
+    The chunk always contains whole frames (senders send record-aligned).
+    Native-accelerated when the codec extension is built."""
+    if _native is not None:
+        return iter(_native.parse_frames(data))
+    return _parse_frames_py(data)
+
+
+def _parse_frames_py(data: bytes) -> Iterator[Tuple[int, bytes]]:
     pos = 0
     while pos + _HDR.size <= len(data):
         length, crc = _HDR.unpack_from(data, pos)
