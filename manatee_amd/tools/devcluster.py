@@ -376,3 +376,61 @@ class DevCluster:
                     await cli.close()
             await asyncio.sleep(0.1)
         raise AssertionError("cluster never became writable: %r" % last_err)
+
+
+# ------------------------------------------------------------------ CLI
+def main(argv=None) -> int:
+    """Dev-cluster launcher — the mkdevsitters + "start all sitters"
+    analogue (ref tools/mkdevsitters, docs/working-on-manatee.md):
+    builds N peers with stepped ports under a directory, starts the
+    embedded ZK + all daemons, prints how to reach them, and runs until
+    interrupted."""
+    import argparse
+    import signal as _signal
+
+    ap = argparse.ArgumentParser(
+        prog="manatee-devcluster",
+        description="Run a local N-peer manatee shard for development")
+    ap.add_argument("-d", "--dir", required=True,
+                    help="base directory for peer state/logs")
+    ap.add_argument("-n", "--peers", type=int, default=3)
+    ap.add_argument("-s", "--shard", default="1.dev")
+    ap.add_argument("--singleton", action="store_true",
+                    help="one peer in one-node-write mode")
+    ap.add_argument("--session-timeout-ms", type=int, default=10000)
+    ns = ap.parse_args(argv)
+
+    async def run():
+        c = DevCluster(ns.dir, n_peers=ns.peers, shard_name=ns.shard,
+                       singleton=ns.singleton or ns.peers == 1,
+                       session_timeout_ms=ns.session_timeout_ms)
+        await c.start()
+        print("zk:     %s" % c.zk_conn_str)
+        print("shard:  %s" % c.shard_path)
+        for p in c.peers:
+            print("peer%d:  db %s:%d  status http://%s:%d  backup "
+                  "http://%s:%d  dir %s"
+                  % (p.index, p.ip, p.pg_port, p.ip, p.status_port,
+                     p.ip, p.backup_port, p.dir))
+        print("adm:    ZK_IPS=%s SHARD=%s bin/manatee-adm pg-status"
+              % (c.zk_conn_str, c.shard_path))
+        try:
+            await c.wait_cluster(lambda s: s.get("primary"),
+                                 timeout_s=120, what="formation")
+            print("cluster formed; Ctrl-C to stop")
+        except AssertionError as exc:
+            print("WARNING: %s" % exc)
+        stop = asyncio.Event()
+        loop = asyncio.get_running_loop()
+        for sig in (_signal.SIGINT, _signal.SIGTERM):
+            loop.add_signal_handler(sig, stop.set)
+        await stop.wait()
+        c.stop()
+        return 0
+
+    return asyncio.run(run())
+
+
+if __name__ == "__main__":
+    import sys as _sys
+    _sys.exit(main())
