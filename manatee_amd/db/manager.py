@@ -243,12 +243,33 @@ class DbManager:
         downstream = cfg.get("downstream")
         self.log.info("transitioning to primary",
                       downstream=(downstream or {}).get("pgUrl"))
+        was_standby = self.engine.current_conf_role() == "standby"
+        onwm = self.one_node_write_mode or downstream is None
+        if was_standby and self.online \
+                and self._proc is not None \
+                and self._proc.returncode is None:
+            # ONLINE promote (the pg_ctl-promote discipline): the running
+            # standby switches role in place on SIGHUP — no process
+            # restart on the failover critical path
+            self.engine.write_conf("primary", read_only=not onwm)
+            self.engine.write_promote_trigger()
+            self._reload_db()
+            if await self._await_promoted(timeout_s=5.0):
+                self.writable = bool(onwm)
+                asyncio.get_running_loop().create_task(
+                    self._background_snapshot())
+                if not onwm:
+                    self._transition_task = \
+                        asyncio.get_running_loop().create_task(
+                            self._wait_for_standby(downstream))
+                return
+            self.log.warn("online promote did not take; falling back to "
+                          "a restart")
         await self._stop_db()
         if not self.engine.initialized():
             await self.store.ensure()
             await self.engine.init_datadir()
         was_standby = self.engine.current_conf_role() == "standby"
-        onwm = self.one_node_write_mode or downstream is None
         # start read-only unless ONWM (ref :1145-1154); sync names are set
         # only after the standby has caught up (ref :1077-1085)
         self.engine.write_conf("primary", read_only=not onwm)
@@ -263,6 +284,20 @@ class DbManager:
         if not onwm:
             self._transition_task = asyncio.get_running_loop().create_task(
                 self._wait_for_standby(downstream))
+
+    async def _await_promoted(self, timeout_s: float = 5.0) -> bool:
+        """Poll until the engine reports role=primary after an online
+        promote reload."""
+        deadline = time.monotonic() + timeout_s
+        while time.monotonic() < deadline:
+            try:
+                status = await self.engine.status()
+                if status.get("role") == "primary":
+                    return True
+            except Exception:
+                pass
+            await asyncio.sleep(0.02)
+        return False
 
     async def _background_snapshot(self) -> None:
         try:

@@ -354,3 +354,42 @@ def test_adm_cli_rebuild_live(cluster_dir):
         finally:
             c.stop()
     run(go(), timeout=300)
+
+
+def test_takeover_promotes_db_online_without_restart(cluster_dir):
+    """The sync's database must be promoted IN PLACE on takeover (the
+    pg_ctl-promote discipline): same db process before and after the
+    failover, timeline bumped."""
+    async def go():
+        c = DevCluster(cluster_dir, n_peers=3, shard_name="1.online")
+        try:
+            await c.start()
+            s = await c.wait_cluster(
+                lambda s: s.get("sync") and len(s.get("async", [])) == 1,
+                timeout_s=60, what="formation")
+            prim = await c.wait_writable(timeout_s=60)
+            sync_peer = c.peer_by_id(s["sync"]["id"])
+            pid_before = sync_peer.db_pid()
+            assert pid_before is not None
+            scli = sync_peer.db_client()
+            st = await scli.status()
+            tl_before = st["timeline"]
+            await scli.close()
+
+            prim.kill9()
+            await c.wait_cluster(
+                lambda s2: s2["generation"] > s["generation"]
+                and s2["primary"]["id"] == sync_peer.id,
+                timeout_s=60, what="sync takeover")
+            await c.wait_writable(timeout_s=60)
+
+            assert sync_peer.db_pid() == pid_before, \
+                "takeover restarted the database process"
+            scli = sync_peer.db_client()
+            st = await scli.status()
+            assert st["role"] == "primary"
+            assert st["timeline"] == tl_before + 1
+            await scli.close()
+        finally:
+            c.stop()
+    run(go())
