@@ -23,6 +23,7 @@ import asyncio
 import json
 import os
 import random
+import re
 import struct
 import time
 from typing import Dict, List, Optional, Set, Tuple
@@ -79,6 +80,9 @@ class _Session:
         self.closed = False
 
 
+_TRAILING_SEQ = re.compile(r"(\d{10})$")
+
+
 def _parent(path: str) -> str:
     i = path.rfind("/")
     return path[:i] if i > 0 else "/"
@@ -108,6 +112,8 @@ class ZkServer:
         self.max_to = max_session_timeout_ms
         self.journal_path = journal_path
         self._journal = None
+        self._journal_entries = 0
+        self.journal_compact_entries = 5000
 
         self.nodes: Dict[str, _Node] = {"/": _Node(b"", 0)}
         self.sessions: Dict[int, _Session] = {}
@@ -160,6 +166,13 @@ class ZkServer:
     def _journal_write(self, op: dict) -> None:
         if self._journal is not None:
             self._journal.write(json.dumps(op, separators=(",", ":")) + "\n")
+            self._journal_entries += 1
+            if self._journal_entries >= self.journal_compact_entries and \
+                    self._journal_entries >= 2 * max(1, len(self.nodes)):
+                try:
+                    self._compact_journal()
+                except OSError as exc:
+                    self.log.error("journal compaction failed", err=exc)
 
     def _replay_journal(self) -> None:
         n = 0
@@ -171,10 +184,20 @@ class ZkServer:
                 op = json.loads(line)
                 try:
                     if op["op"] == "create":
+                        # journaled paths are FINAL (already sequenced):
+                        # always replay as plain persistent creates, then
+                        # restore the parent's sequence counter so new
+                        # sequential nodes continue after the replayed ones
                         self._do_create(op["path"],
                                         bytes.fromhex(op["data"]),
-                                        op["flags"] & ~jute.EPHEMERAL, 0,
+                                        jute.PERSISTENT, 0,
                                         journal=False)
+                        m = _TRAILING_SEQ.search(op["path"])
+                        if m:
+                            parent = self.nodes.get(_parent(op["path"]))
+                            if parent is not None:
+                                parent.next_seq = max(
+                                    parent.next_seq, int(m.group(1)) + 1)
                     elif op["op"] == "setData":
                         self._do_set_data(op["path"],
                                           bytes.fromhex(op["data"]), -1,
@@ -184,7 +207,33 @@ class ZkServer:
                 except ZkError:
                     pass
                 n += 1
+        self._journal_entries = n
         self.log.info("journal replayed", entries=n)
+
+    def _compact_journal(self) -> None:
+        """Rewrite the journal as a snapshot of the current persistent
+        tree (one create per node) — bounds the append-only journal,
+        which otherwise grows with every cluster-state write forever."""
+        tmp = self.journal_path + ".tmp"
+        count = 0
+        with open(tmp, "w") as f:
+            for path in sorted(self.nodes):
+                node = self.nodes[path]
+                if path == "/" or node.ephemeral_owner:
+                    continue
+                f.write(json.dumps(
+                    {"op": "create", "path": path,
+                     "data": node.data.hex(), "flags": jute.PERSISTENT},
+                    separators=(",", ":")) + "\n")
+                count += 1
+            f.flush()
+            os.fsync(f.fileno())
+        if self._journal is not None:
+            self._journal.close()
+        os.replace(tmp, self.journal_path)
+        self._journal = open(self.journal_path, "a", buffering=1)
+        self._journal_entries = count
+        self.log.info("journal compacted", nodes=count)
 
     # ------------------------------------------------------- session sweeper
     async def _sweep_sessions(self) -> None:

@@ -251,3 +251,54 @@ def test_journal_restart(tmp_path):
 
     run(phase1())
     run(phase2())
+
+
+def test_journal_sequential_replay_and_compaction(tmp_path):
+    """Persistent-sequential nodes must replay with their FINAL names
+    (no re-sequencing) and the parent's counter must continue after
+    them; the journal compacts to a tree snapshot once it outgrows the
+    node count and the compacted journal replays identically."""
+    from manatee_amd.coord import jute
+
+    j = str(tmp_path / "zk.jsonl")
+
+    async def go():
+        srv = ZkServer(journal_path=j)
+        await srv.start()
+        cli = ZkClient(srv.conn_str)
+        await cli.connect()
+        await cli.mkdirp("/hist")
+        await cli.create("/hist/gen-", b"a", jute.PERSISTENT_SEQUENTIAL)
+        await cli.create("/hist/gen-", b"b", jute.PERSISTENT_SEQUENTIAL)
+        await cli.close()
+        await srv.stop()
+
+        srv2 = ZkServer(host=srv.host, port=srv.port, journal_path=j)
+        await srv2.start()
+        cli2 = ZkClient(srv2.conn_str)
+        await cli2.connect()
+        ch, _ = await cli2.get_children("/hist")
+        assert sorted(ch) == ["gen-0000000000", "gen-0000000001"], ch
+        p3 = await cli2.create("/hist/gen-", b"c",
+                               jute.PERSISTENT_SEQUENTIAL)
+        assert p3 == "/hist/gen-0000000002"
+
+        srv2.journal_compact_entries = 10
+        for i in range(30):
+            await cli2.set_data("/hist/gen-0000000000", b"x%d" % i)
+        with open(j) as f:
+            assert sum(1 for _ in f) < 30, "journal did not compact"
+        await cli2.close()
+        await srv2.stop()
+
+        srv3 = ZkServer(host=srv.host, port=srv.port, journal_path=j)
+        await srv3.start()
+        cli3 = ZkClient(srv3.conn_str)
+        await cli3.connect()
+        data, _ = await cli3.get_data("/hist/gen-0000000000")
+        assert data == b"x29"
+        ch, _ = await cli3.get_children("/hist")
+        assert len(ch) == 3
+        await cli3.close()
+        await srv3.stop()
+    run(go())
