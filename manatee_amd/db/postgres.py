@@ -1,0 +1,423 @@
+"""PostgresEngine — run real PostgreSQL under the sitter.
+
+The engine-interface implementation of the reference's PostgresMgr
+database specifics (ref lib/postgresMgr.js):
+
+- multi-version support 9.2 / 9.6 / 12 with the wal/xlog and
+  lsn/location name translations (ref resolveWalTranslations :649-677),
+  per-version binary dirs via a ``current`` symlink and version
+  metadata persisted in ``manatee-config.json`` on the dataset
+  (ref resolveVersionedPaths :569-634, getVersionInfo :446-510);
+- operator tunables from ``pg_overrides.json`` merged
+  common → major → full-version (ref getTunables :527-560);
+- conf regeneration on every transition (never edited in place, custom
+  keys in a live conf are lost — ref :2282-2336): standbys get
+  ``primary_conninfo`` (+ ``recovery.conf`` with ``standby_mode=on``
+  before PG 12, ``standby.signal`` from 12 on — ref
+  _updateUpstreamConf :2188-2274) and ``synchronous_commit=off``;
+  primaries get ``synchronous_standby_names`` only after catch-up;
+- health/status/replication probes over the native wire client
+  (pgwire), one serialized connection (ref query queue :1990-2172).
+
+Config keys (``postgresMgrCfg``, ref CONFIG_SCHEMA :60-116):
+``versions {"9.2": "9.2.4", ...}``, ``defaultVersion``, ``pgBaseDir``,
+``postgresConfDir``, ``postgresConfFile``, ``recoveryConfFile``,
+``hbaConfFile``, ``tunablesFile``, ``dbUser``, ``pgConnectTimeout``.
+"""
+
+from __future__ import annotations
+
+import asyncio
+import json
+import os
+import shutil
+import time
+from typing import Dict, List, Optional
+
+from ..common import confparser
+from ..common import procutil
+from ..common.logging import Logger, null_logger
+from ..common.lsn import pg_strip_minor
+from .engine import Engine
+from .pgwire import PgClient, PgError
+
+DATA_CONF_NAME = "manatee-config.json"
+PROMOTE_TRIGGER = "promote"
+
+# fixed template values (ref etc/postgresql.conf; SURVEY.md §6 durability
+# floor): hot-standby WAL, remote_write sync commit
+BASE_CONF = {
+    "wal_level": "hot_standby",
+    "hot_standby": "on",
+    "synchronous_commit": "remote_write",
+    "full_page_writes": "off",
+    "max_wal_senders": "10",
+    "wal_keep_segments": "64",
+}
+
+
+def resolve_wal_translations(major: str) -> dict:
+    """PG 10 renamed xlog→wal and location→lsn
+    (ref resolveWalTranslations lib/postgresMgr.js:649-677)."""
+    translations = {
+        "lsn": "lsn",
+        "wal": "wal",
+        "queries": {
+            "current_lsn": "SELECT pg_current_wal_lsn() as loc;",
+            "last_replay_lsn": "SELECT pg_last_wal_replay_lsn() as loc;",
+        },
+    }
+    if major in ("9.2", "9.6"):
+        translations["lsn"] = "location"
+        translations["wal"] = "xlog"
+        translations["queries"]["current_lsn"] = \
+            "SELECT pg_current_xlog_location() as loc;"
+        translations["queries"]["last_replay_lsn"] = \
+            "SELECT pg_last_xlog_replay_location() as loc;"
+    return translations
+
+
+def get_tunables(tunables: dict, version: str, major: str) -> Dict[str, str]:
+    """Layered overrides, most specific last
+    (ref getTunables lib/postgresMgr.js:527-560)."""
+    options: Dict[str, str] = {"synchronous_commit": "remote_write"}
+    for layer in ("common", major, version):
+        for k, v in (tunables.get(layer) or {}).items():
+            options[k] = str(v)
+    return options
+
+
+def get_version_info(data_dir: str, data_conf: str,
+                     versions: Dict[str, str],
+                     default_version: str) -> dict:
+    """{initialized, current} from manatee-config.json + PG_VERSION
+    (ref getVersionInfo lib/postgresMgr.js:446-510)."""
+    pgc = None
+    try:
+        with open(data_conf) as f:
+            pgc = json.load(f)
+    except FileNotFoundError:
+        pass
+    curver = None
+    try:
+        with open(os.path.join(data_dir, "PG_VERSION")) as f:
+            curver = f.read().strip()
+    except FileNotFoundError:
+        pass
+
+    if pgc is None:
+        if curver is None:
+            full = versions[default_version]
+            return {"initialized": full, "current": full}
+        # dataset predates manatee-config.json ⇒ it is a 9.2 dataset
+        if curver != "9.2":
+            raise ValueError("dataset without %s must be 9.2, found %r"
+                             % (DATA_CONF_NAME, curver))
+        return {"initialized": "9.2.4", "current": versions["9.2"]}
+
+    major = pg_strip_minor(pgc["current"])
+    current = versions.get(major)
+    if current is None:
+        raise ValueError("no configured binaries for major %r" % major)
+    if pgc["current"] != current:
+        raise ValueError("dataset patch version %r != configured %r"
+                         % (pgc["current"], current))
+    if curver is not None and curver != major:
+        raise ValueError("PG_VERSION %r != current major %r"
+                         % (curver, major))
+    return {"initialized": pgc["initialized"], "current": current}
+
+
+class PostgresEngine(Engine):
+    name = "postgres"
+
+    def __init__(self, data_dir: str, ip: str, port: int, peer_name: str,
+                 cfg: Optional[dict] = None, log: Optional[Logger] = None):
+        cfg = cfg or {}
+        self.data_dir = data_dir
+        self.ip = ip
+        self.port = port
+        self.peer_name = peer_name
+        self.log = (log or null_logger()).child(component="PostgresEngine")
+        self.versions: Dict[str, str] = cfg.get(
+            "versions", {"12": "12.0"})
+        self.default_version = cfg.get("defaultVersion",
+                                       sorted(self.versions)[-1])
+        self.pg_base_dir = cfg.get("pgBaseDir", "/opt/postgresql")
+        self.conf_dir = cfg.get("postgresConfDir")   # template dir per major
+        self.postgres_conf_file = cfg.get("postgresConfFile",
+                                          "postgresql.conf")
+        self.recovery_conf_file = cfg.get("recoveryConfFile",
+                                          "recovery.conf")
+        self.hba_conf_file = cfg.get("hbaConfFile", "pg_hba.conf")
+        self.tunables_file = cfg.get("tunablesFile")
+        self.db_user = cfg.get("dbUser", "postgres")
+        self.connect_timeout_s = cfg.get("pgConnectTimeout", 60)
+        self.dataset_dir = os.path.dirname(os.path.abspath(data_dir))
+        self.data_conf = os.path.join(self.dataset_dir, DATA_CONF_NAME)
+        # resolved by resolve_versioned_paths():
+        self.current_version: Optional[str] = None
+        self.major: Optional[str] = None
+        self.bin_dir: Optional[str] = None
+        self.wal_translations: dict = resolve_wal_translations("12")
+        self.tunables: Dict[str, str] = {}
+        self.uses_signal = True
+        self._client: Optional[PgClient] = None
+
+    # ----------------------------------------------------- version plumbing
+    def resolve_versioned_paths(self) -> dict:
+        """ref resolveVersionedPaths :569-634."""
+        verinfo = get_version_info(self.data_dir, self.data_conf,
+                                   self.versions, self.default_version)
+        os.makedirs(self.dataset_dir, exist_ok=True)
+        with open(self.data_conf, "w") as f:
+            json.dump(verinfo, f)
+        version = verinfo["current"]
+        self.current_version = version
+        self.major = pg_strip_minor(version)
+        vers_dir = os.path.join(self.pg_base_dir, version)
+        curr_link = os.path.join(self.pg_base_dir, "current")
+        try:
+            if os.path.islink(curr_link):
+                os.unlink(curr_link)
+            os.symlink(vers_dir, curr_link)
+        except OSError:
+            pass   # base dir may be read-only/absent in dev
+        self.bin_dir = os.path.join(vers_dir, "bin")
+        self.uses_signal = float(self.major) >= 12
+        self.wal_translations = resolve_wal_translations(self.major)
+        tunables = {}
+        if self.tunables_file:
+            with open(self.tunables_file) as f:
+                tunables = json.load(f)
+        self.tunables = get_tunables(tunables, version, self.major)
+        return verinfo
+
+    def _template_conf(self) -> Dict[str, str]:
+        if self.conf_dir and self.major:
+            path = os.path.join(self.conf_dir, self.major,
+                                self.postgres_conf_file)
+            if os.path.exists(path):
+                return confparser.read(path)
+        return dict(BASE_CONF)
+
+    # ------------------------------------------------------------ interface
+    def initialized(self) -> bool:
+        return os.path.exists(os.path.join(self.data_dir, "PG_VERSION"))
+
+    async def init_datadir(self) -> None:
+        """initdb via fork-exec (ref _prepareDatabase :1806-1987)."""
+        self.resolve_versioned_paths()
+        os.makedirs(self.data_dir, exist_ok=True)
+        os.chmod(self.data_dir, 0o700)
+        initdb = os.path.join(self.bin_dir, "initdb")
+        argv = [initdb, "-D", self.data_dir, "-E", "UTF8"]
+        if os.geteuid() == 0:
+            import pwd
+            try:
+                pw = pwd.getpwnam(self.db_user)
+                procutil.chown_r(self.data_dir, pw.pw_uid, pw.pw_gid)
+                argv = ["sudo", "-u", self.db_user] + argv
+            except KeyError:
+                self.log.warn("db user missing; running initdb as root",
+                              user=self.db_user)
+        await procutil.run_async(argv, env=procutil.SCRUBBED_ENV,
+                                 timeout=300)
+        self._install_hba()
+        self.log.info("initdb complete", dataDir=self.data_dir,
+                      version=self.current_version)
+
+    def _install_hba(self) -> None:
+        """Install pg_hba (template, or permissive trust defaults)."""
+        dest = os.path.join(self.data_dir, "pg_hba.conf")
+        if self.conf_dir and self.major:
+            src = os.path.join(self.conf_dir, self.major,
+                               self.hba_conf_file)
+            if os.path.exists(src):
+                shutil.copyfile(src, dest)
+                return
+        with open(dest, "w") as f:
+            f.write("# generated by manatee-amd\n"
+                    "local   all         all                     trust\n"
+                    "host    all         all   0.0.0.0/0         trust\n"
+                    "host    replication all   0.0.0.0/0         trust\n")
+
+    def _conf_path(self) -> str:
+        return os.path.join(self.data_dir, self.postgres_conf_file)
+
+    def _recovery_path(self) -> str:
+        return os.path.join(self.data_dir, self.recovery_conf_file)
+
+    def _signal_path(self) -> str:
+        return os.path.join(self.data_dir, "standby.signal")
+
+    def _trigger_path(self) -> str:
+        return os.path.join(self.dataset_dir, PROMOTE_TRIGGER)
+
+    def current_conf_role(self) -> Optional[str]:
+        if os.path.exists(self._recovery_path()) or \
+                os.path.exists(self._signal_path()):
+            return "standby"
+        if os.path.exists(self._conf_path()):
+            return "primary"
+        return None
+
+    def _primary_conninfo(self, upstream_url: str) -> str:
+        """ref PRIMARY_CONNINFO_STR lib/postgresMgr.js:167-173."""
+        from .engine import url_to_hostport
+        host, port = url_to_hostport(upstream_url)
+        return ("'host=%s port=%d user=%s application_name=%s "
+                "connect_timeout=%d'"
+                % (host, port, self.db_user, self.peer_name,
+                   self.connect_timeout_s))
+
+    def write_conf(self, role: str, upstream_url: Optional[str] = None,
+                   sync_name: Optional[str] = None,
+                   read_only: bool = False) -> None:
+        """Regenerate postgresql.conf (+ recovery.conf / standby.signal)
+        from the template — never edit in place (ref :2282-2336)."""
+        self.resolve_versioned_paths()
+        conf = self._template_conf()
+        conf.update(self.tunables)
+        conf["listen_addresses"] = "'%s'" % self.ip
+        conf["port"] = str(self.port)
+        if role == "primary":
+            conf["default_transaction_read_only"] = \
+                "on" if read_only else "off"
+            if sync_name:
+                conf["synchronous_standby_names"] = "'%s'" % sync_name
+            # leaving standby mode (ref _primary :1135-1144)
+            for stale in (self._recovery_path(), self._signal_path()):
+                try:
+                    os.unlink(stale)
+                except FileNotFoundError:
+                    pass
+        else:
+            assert upstream_url, "standby requires an upstream"
+            conninfo = self._primary_conninfo(upstream_url)
+            # standbys never gate their own commits (ref SYNCHRONOUS_COMMIT
+            # 'off' in _updateUpstreamConf :2209)
+            conf["synchronous_commit"] = "off"
+            if self.uses_signal:
+                conf["primary_conninfo"] = conninfo
+                conf["recovery_target_timeline"] = "'latest'"
+                conf["promote_trigger_file"] = "'%s'" % self._trigger_path()
+                open(self._signal_path(), "w").close()
+            else:
+                rec = {
+                    "standby_mode": "on",
+                    "primary_conninfo": conninfo,
+                    "recovery_target_timeline": "'latest'",
+                    "trigger_file": "'%s'" % self._trigger_path(),
+                }
+                confparser.write(self._recovery_path(), rec)
+        confparser.write(self._conf_path(), conf)
+
+    def write_promote_trigger(self) -> None:
+        open(self._trigger_path(), "w").close()
+
+    def post_restore_fixup(self) -> None:
+        for name in ("postmaster.pid", "recovery.done", "db_child.pid"):
+            try:
+                os.unlink(os.path.join(self.data_dir, name))
+            except FileNotFoundError:
+                pass
+
+    def spawn_argv(self) -> List[str]:
+        if self.bin_dir is None:
+            self.resolve_versioned_paths()
+        return [os.path.join(self.bin_dir, "postgres"), "-D", self.data_dir]
+
+    # --------------------------------------------------------------- probes
+    def _cli(self) -> PgClient:
+        if self._client is None or not self._client.connected:
+            self._client = PgClient(self.ip, self.port, self.db_user)
+        return self._client
+
+    async def _connected_cli(self) -> PgClient:
+        cli = self._cli()
+        if not cli.connected:
+            await cli.connect()
+        return cli
+
+    async def ping(self, timeout_s: float = 5.0) -> bool:
+        """ref health check 'select current_time' :1550-1626."""
+        try:
+            cli = await asyncio.wait_for(self._connected_cli(), timeout_s)
+            await cli.query("SELECT current_time;", timeout_s=timeout_s)
+            return True
+        except (PgError, OSError, asyncio.TimeoutError, ConnectionError):
+            await self.close()
+            return False
+
+    async def _is_in_recovery(self, cli: PgClient) -> bool:
+        res = await cli.query("SELECT pg_is_in_recovery() as r;")
+        return res.rows and res.rows[0][0] in ("t", "true", "True")
+
+    async def xlog(self) -> str:
+        """ref getXLogLocation :878-899 — role picks the query."""
+        cli = await self._connected_cli()
+        q = self.wal_translations["queries"]
+        if await self._is_in_recovery(cli):
+            res = await cli.query(q["last_replay_lsn"])
+        else:
+            res = await cli.query(q["current_lsn"])
+        return res.rows[0][0]
+
+    async def status(self) -> dict:
+        cli = await self._connected_cli()
+        in_recovery = await self._is_in_recovery(cli)
+        lsn_word = self.wal_translations["lsn"]
+        repl = []
+        res = await cli.query("SELECT * FROM pg_stat_replication;")
+        for row in res.dicts():
+            repl.append({
+                "application_name": row.get("application_name"),
+                "state": row.get("state"),
+                "sync_state": row.get("sync_state"),
+                "sent_lsn": row.get("sent_" + lsn_word),
+                "write_lsn": row.get("write_" + lsn_word),
+                "flush_lsn": row.get("flush_" + lsn_word),
+                "replay_lsn": row.get("replay_" + lsn_word),
+            })
+        q = self.wal_translations["queries"]
+        cur = await cli.query(q["last_replay_lsn" if in_recovery
+                                else "current_lsn"])
+        lrt = None
+        if in_recovery:
+            r = await cli.query(
+                "SELECT extract(epoch from "
+                "pg_last_xact_replay_timestamp()) as t;")
+            if r.rows and r.rows[0][0] is not None:
+                lrt = float(r.rows[0][0])
+        return {
+            "ok": True,
+            "role": "standby" if in_recovery else "primary",
+            "current_lsn": cur.rows[0][0],
+            "replay_lsn": cur.rows[0][0] if in_recovery else None,
+            "last_replay_time": lrt,
+            "replication": repl,
+        }
+
+    async def check_repl(self, standby_name: str) -> dict:
+        """ref _checkReplStatus :2478-2556 — caught up when
+        sent == flush for the named standby."""
+        st = await self.status()
+        row = next((r for r in st.get("replication", [])
+                    if r["application_name"] == standby_name), None)
+        if row is None:
+            return {"connected": False, "caught_up": False}
+        caught_up = bool(row["sent_lsn"]) and \
+            row["sent_lsn"] == row["flush_lsn"]
+        return {"connected": True, "sync_state": row["sync_state"],
+                "sent_lsn": row["sent_lsn"], "write_lsn": row["write_lsn"],
+                "flush_lsn": row["flush_lsn"],
+                "replay_lsn": row["replay_lsn"], "caught_up": caught_up}
+
+    async def close(self) -> None:
+        if self._client is not None:
+            try:
+                await self._client.close()
+            except Exception:
+                pass
+            self._client = None
