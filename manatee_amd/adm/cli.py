@@ -390,8 +390,28 @@ async def cmd_check_lock(ns) -> int:
 
 
 async def cmd_promote(ns) -> int:
-    shard = _need(ns, "shard", "SHARD", "-s/--shard")
     role = ns.role or "sync"
+
+    # pre-flight: cluster health + replication lag (ref promote
+    # :1726-1850 — refuse when the cluster has issues or any peer lags
+    # more than lagToIgnore, unless forced with -y)
+    try:
+        cd = await _details(ns)
+    except Exception as exc:
+        return _fail("cannot load cluster state: %s" % exc)
+    warnings = ["cluster error: %s" % e for e in cd.errors]
+    warnings += ["cluster warning: %s" % w for w in cd.warnings]
+    for pd in cd.peers.values():
+        if pd.lag_s is not None and pd.lag_s > ns.lag_to_ignore:
+            warnings.append('"%s" has %ds of lag behind its upstream '
+                            "peer" % (pd.label, int(pd.lag_s)))
+    if warnings and not ns.yes:
+        for w in warnings:
+            print("warning: %s" % w, file=sys.stderr)
+        return _fail("refusing to promote with outstanding warnings "
+                     "(use -y to override, or -l to raise the lag "
+                     "threshold)")
+    shard = _need(ns, "shard", "SHARD", "-s/--shard")
 
     async def go(zk):
         try:
@@ -606,6 +626,11 @@ def _mk_parser() -> argparse.ArgumentParser:
     sp.add_argument("-n", "--zonename")
     sp.add_argument("--role", choices=("sync", "async"))
     sp.add_argument("--asyncIndex", dest="async_index", type=int)
+    sp.add_argument("-l", "--lagToIgnore", dest="lag_to_ignore",
+                    type=int, default=60,
+                    help="max acceptable replay lag in seconds")
+    sp.add_argument("-y", "--yes", action="store_true",
+                    help="promote despite warnings")
 
     add("clear-promote", cmd_clear_promote,
         help="remove a pending promote request")

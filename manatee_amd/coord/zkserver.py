@@ -237,8 +237,10 @@ class ZkServer:
 
     # ------------------------------------------------------- session sweeper
     async def _sweep_sessions(self) -> None:
+        # sweep at tick/4 (min 20 ms): expiry granularity is pure added
+        # failover-detection latency and the scan is O(sessions), tiny
         while True:
-            await asyncio.sleep(self.tick_ms / 1000.0)
+            await asyncio.sleep(max(0.02, self.tick_ms / 4000.0))
             now = time.monotonic()
             for sess in list(self.sessions.values()):
                 if sess.closed:

@@ -168,6 +168,36 @@ class DevPeer:
         self.sitter_proc = None
         self.backup_proc = None
 
+    def pause(self) -> None:
+        """SIGSTOP the whole peer (sitter pg + db pg) — the network-
+        partition analogue on one host: processes stay alive but stop
+        responding, the ZK session expires, and on resume() the peer
+        discovers the cluster moved on without it."""
+        pids = [p.pid for p in (self.sitter_proc, self.backup_proc)
+                if p is not None and p.poll() is None]
+        pids += self.db_pids()
+        for pid in pids:
+            try:
+                os.killpg(pid, signal.SIGSTOP)
+            except (ProcessLookupError, PermissionError):
+                try:
+                    os.kill(pid, signal.SIGSTOP)
+                except ProcessLookupError:
+                    pass
+
+    def resume(self) -> None:
+        pids = [p.pid for p in (self.sitter_proc, self.backup_proc)
+                if p is not None and p.poll() is None]
+        pids += self.db_pids()
+        for pid in pids:
+            try:
+                os.killpg(pid, signal.SIGCONT)
+            except (ProcessLookupError, PermissionError):
+                try:
+                    os.kill(pid, signal.SIGCONT)
+                except ProcessLookupError:
+                    pass
+
     def kill_db_only(self) -> None:
         """SIGKILL only the database child (the sitter must notice and
         restart it)."""

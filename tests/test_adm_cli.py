@@ -171,6 +171,9 @@ EXTRA_COMMANDS = {        # flag permutations, run for normalOk only
         ["pg-status", "-r", "bogus"],
         ["peers", "-o", "nonexistent"],
         ["zk-state"],
+        # lag pre-flight: the async lags 5m42s > 60s -> refuse (exit 1)
+        # before any ZK access (ref promote lag check lib/adm.js:1833-1843)
+        ["promote", "-i", "10.0.0.3:5432:5434", "--role", "async"],
     ],
 }
 

@@ -393,3 +393,70 @@ def test_takeover_promotes_db_online_without_restart(cluster_dir):
         finally:
             c.stop()
     run(go())
+
+
+def test_partitioned_primary_no_split_brain(cluster_dir):
+    """Partition analogue (docs/test-plan.md network-partition tier):
+    SIGSTOP the whole primary peer.  Its ZK session expires and the
+    sync takes over.  The frozen old primary must NOT be able to
+    acknowledge writes when it wakes (its sync standby is gone — the
+    remote_write gate blocks), and once its sitter resumes it must
+    observe that it is deposed and stop serving as primary."""
+    async def go():
+        c = DevCluster(cluster_dir, n_peers=3, shard_name="1.part")
+        try:
+            await c.start()
+            s = await c.wait_cluster(
+                lambda s: s.get("sync") and len(s.get("async", [])) == 1,
+                timeout_s=60, what="formation")
+            prim = await c.wait_writable(timeout_s=60)
+            gen = s["generation"]
+
+            prim.pause()
+            s2 = await c.wait_cluster(
+                lambda s2: s2["generation"] > gen
+                and s2["primary"]["id"] == s["sync"]["id"],
+                timeout_s=60, what="takeover around partitioned primary")
+            assert any(d["id"] == prim.id for d in s2["deposed"])
+            new_prim = await c.wait_writable(timeout_s=60)
+
+            # wake the old primary: a write against it must NOT be
+            # acknowledged (sync gate) and the peer must demote itself
+            prim.resume()
+            cli = prim.db_client()
+            acked = False
+            try:
+                await cli.put("split-brain", 1, timeout_s=2.0)
+                acked = True
+            except Exception:
+                pass
+            await cli.close()
+            assert not acked, \
+                "deposed ex-primary acknowledged a write (split brain!)"
+
+            # the resumed sitter sees it is deposed and stops its db
+            deadline = time.monotonic() + 60
+            demoted = False
+            while time.monotonic() < deadline:
+                cli = prim.db_client()
+                try:
+                    st = await asyncio.wait_for(cli.status(), 1.0)
+                    if st.get("role") != "primary":
+                        demoted = True
+                except Exception:
+                    demoted = True   # db stopped entirely — also fine
+                finally:
+                    await cli.close()
+                if demoted:
+                    break
+                await asyncio.sleep(0.5)
+            assert demoted, "resumed deposed primary still acts as primary"
+
+            # the new primary still works and never lost the write path
+            cli = new_prim.db_client()
+            await cli.put("after-partition", 2)
+            assert await cli.get("after-partition") == 2
+            await cli.close()
+        finally:
+            c.stop()
+    run(go())
