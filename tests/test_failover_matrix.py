@@ -460,3 +460,78 @@ def test_partitioned_primary_no_split_brain(cluster_dir):
         finally:
             c.stop()
     run(go())
+
+
+def test_5peer_cascading_chain_and_middle_death(cluster_dir):
+    """Five peers form primary -> sync -> async0 -> async1 -> async2
+    with each standby following its chain predecessor (cascading
+    replication, ref docs/user-guide.md:69-90,258-266); data reaches the
+    chain tail; killing a middle async re-slaves its successor."""
+    async def go():
+        c = DevCluster(cluster_dir, n_peers=5, shard_name="1.chain")
+        try:
+            await c.start()
+            s = await c.wait_cluster(
+                lambda s: s.get("sync") and len(s.get("async", [])) == 3,
+                timeout_s=90, what="5-peer formation")
+            prim = await c.wait_writable(timeout_s=90)
+
+            # verify the replication edges: each peer's downstream row
+            # names its chain successor
+            chain = [s["primary"], s["sync"]] + s["async"]
+            for up, down in zip(chain[:-1], chain[1:]):
+                peer = c.peer_by_id(up["id"])
+                cli = peer.db_client()
+                deadline = time.monotonic() + 30
+                ok = False
+                while time.monotonic() < deadline and not ok:
+                    st = await cli.status()
+                    ok = any(r["application_name"] == down["id"]
+                             and r["state"] == "streaming"
+                             for r in st.get("replication", []))
+                    if not ok:
+                        await asyncio.sleep(0.2)
+                await cli.close()
+                assert ok, "no stream %s -> %s" % (up["id"], down["id"])
+
+            # a write reaches the chain tail
+            cli = prim.db_client()
+            await cli.put("chain", "deep")
+            await cli.close()
+            tail = c.peer_by_id(s["async"][-1]["id"])
+            tcli = tail.db_client()
+            deadline = time.monotonic() + 30
+            while time.monotonic() < deadline:
+                if await tcli.get("chain") == "deep":
+                    break
+                await asyncio.sleep(0.1)
+            assert await tcli.get("chain") == "deep"
+            await tcli.close()
+
+            # kill the middle async: successor re-slaves, no gen bump
+            gen = s["generation"]
+            mid = c.peer_by_id(s["async"][0]["id"])
+            succ_id = s["async"][1]["id"]
+            mid.kill9()
+            s2 = await c.wait_cluster(
+                lambda s2: len(s2["async"]) == 2
+                and s2["async"][0]["id"] == succ_id,
+                timeout_s=60, what="middle async removal")
+            assert s2["generation"] == gen
+            # the promoted-up async now streams from the sync
+            sync_peer = c.peer_by_id(s["sync"]["id"])
+            scli = sync_peer.db_client()
+            deadline = time.monotonic() + 30
+            ok = False
+            while time.monotonic() < deadline and not ok:
+                st = await scli.status()
+                ok = any(r["application_name"] == succ_id
+                         for r in st.get("replication", []))
+                if not ok:
+                    await asyncio.sleep(0.2)
+            await scli.close()
+            assert ok, "successor did not re-slave to the sync"
+            await c.wait_writable(timeout_s=30)
+        finally:
+            c.stop()
+    run(go())
