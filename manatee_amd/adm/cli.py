@@ -530,12 +530,17 @@ async def cmd_rebuild(ns) -> int:
             return 0
         proc = await asyncio.create_subprocess_shell(ns.start_cmd)
         await proc.wait()
-        # watch the peer's status server until it is back as sync/async
-        # (ref :1550-1678; restore retry budget RESTORE_ATTEMPTS)
+        # watch the peer's /restore progress + the cluster state until it
+        # is back as sync/async (ref :1550-1678; ≤RESTORE_ATTEMPTS
+        # restore failures before giving up)
+        from ..common.httpd import http_request
         status_url = "http://%s:%d" % (cfg["ip"], cfg["postgresPort"] + 1)
-        print("waiting for %s to restore and rejoin (watch %s/restore)"
+        print("waiting for %s to restore and rejoin (watching %s/restore)"
               % (peer_id, status_url))
         deadline = time.monotonic() + ns.timeout
+        last_pct = None
+        failures = 0
+        restore_active = False
         while time.monotonic() < deadline:
             cur, _ = await adm.get_state(zk, shard)
             if cur is not None:
@@ -545,6 +550,32 @@ async def cmd_rebuild(ns) -> int:
                 if peer_id in ids:
                     print("peer %s rejoined the cluster" % peer_id)
                     return 0
+            try:
+                code, body = await http_request(status_url + "/restore",
+                                                timeout_s=2)
+                if code == 200 and isinstance(body, dict):
+                    if body.get("active"):
+                        restore_active = True
+                        size = body.get("size") or 0
+                        if size:
+                            pct = int(100 * (body.get("completed") or 0)
+                                      / size)
+                            if pct != last_pct:
+                                print("restore: %3d%% (%d/%d bytes)"
+                                      % (pct, body.get("completed") or 0,
+                                         size))
+                                last_pct = pct
+                    elif restore_active and body.get("failed"):
+                        restore_active = False
+                        failures += 1
+                        print("restore attempt failed (%d/%d): %s"
+                              % (failures, RESTORE_ATTEMPTS,
+                                 body.get("error")), file=sys.stderr)
+                        if failures >= RESTORE_ATTEMPTS:
+                            return _fail("giving up after %d failed "
+                                         "restore attempts" % failures)
+            except Exception:
+                pass   # status server not up yet
             await asyncio.sleep(1.0)
         return _fail("peer did not rejoin within %ds" % ns.timeout)
     return await _with_zk(ns, go)
