@@ -255,10 +255,16 @@ class DevCluster:
         self.zk_proc: Optional[subprocess.Popen] = None
         self.peers: List[DevPeer] = []
         # pick a base below the ephemeral range so client sockets can never
-        # collide with peer listen ports
+        # collide with peer listen ports; partition the space by RANK so
+        # concurrent torchrun ranks (one shard per rank) can never race
+        # each other into the same block
         lo = max(10000, _EPHEMERAL_LOW - 22000)
+        rank = int(os.environ.get("RANK", "0"))
+        world = max(1, int(os.environ.get("WORLD_SIZE", "1")))
+        span = max(400, 18000 // world)
+        slot = lo + (rank % world) * span
         self._next_base_port = base_port or \
-            (lo + (os.getpid() * 131) % 18000) // 10 * 10
+            (slot + (os.getpid() * 131) % max(1, span - 60)) // 10 * 10
         zk_peer = self.add_peer_config()   # reserve a port block for ZK
         self.peers.clear()
         self.zk_port = zk_peer.pg_port
