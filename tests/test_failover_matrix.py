@@ -535,3 +535,37 @@ def test_5peer_cascading_chain_and_middle_death(cluster_dir):
         finally:
             c.stop()
     run(go())
+
+
+def test_metrics_endpoint(cluster_dir):
+    """GET /metrics serves Prometheus text with role/generation/health/
+    replication series (observability beyond the reference, which has
+    no metrics endpoint — SURVEY.md §5.5)."""
+    async def go():
+        c = DevCluster(cluster_dir, n_peers=3, shard_name="1.metrics")
+        try:
+            await c.start()
+            s = await c.wait_cluster(
+                lambda s: s.get("sync") and len(s.get("async", [])) == 1,
+                timeout_s=60, what="formation")
+            prim = await c.wait_writable(timeout_s=60)
+            code, text = await prim.http_status("/metrics")
+            assert code == 200
+            assert isinstance(text, str)
+            assert "manatee_role 0" in text
+            assert "manatee_db_writable 1" in text
+            assert "manatee_db_healthy 1" in text
+            assert "manatee_generation 1" in text
+            assert "manatee_cluster_frozen 0" in text
+            assert "manatee_wal_lsn_bytes" in text
+            assert 'manatee_replication_unflushed_bytes{downstream="%s"' \
+                % s["sync"]["id"] in text
+
+            apeer = c.peer_by_id(s["async"][0]["id"])
+            code, text = await apeer.http_status("/metrics")
+            assert code == 200
+            assert "manatee_role 2" in text
+            assert "manatee_db_writable 0" in text
+        finally:
+            c.stop()
+    run(go())
