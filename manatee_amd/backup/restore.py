@@ -104,9 +104,13 @@ class RestoreClient:
                                                 self.listen_port)
             port = server.sockets[0].getsockname()[1]
             try:
+                # short timeout: an unresponsive restore peer (paused,
+                # partitioned, dying) must fail fast so the FSM can
+                # re-evaluate with a fresh restorePeer instead of
+                # stalling a whole evaluation cycle
                 status, resp = await http_request(
                     backup_url.rstrip("/") + "/backup", "POST",
-                    {"host": self.listen_ip, "port": port})
+                    {"host": self.listen_ip, "port": port}, timeout_s=10.0)
                 if status != 200 or not isinstance(resp, dict):
                     raise RestoreError("backup request refused: %s %r"
                                        % (status, resp))
@@ -117,7 +121,8 @@ class RestoreClient:
                 while True:
                     if recv_done.done():
                         recv_done.result()  # raises on stream failure
-                    jstatus, job = await http_request(job_url)
+                    jstatus, job = await http_request(job_url,
+                                                      timeout_s=10.0)
                     if jstatus == 200 and isinstance(job, dict):
                         ro.size = job.get("size", 0)
                         if job.get("failed"):

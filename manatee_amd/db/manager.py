@@ -254,7 +254,7 @@ class DbManager:
             self.engine.write_conf("primary", read_only=not onwm)
             self.engine.write_promote_trigger()
             self._reload_db()
-            if await self._await_promoted(timeout_s=5.0):
+            if await self._await_promoted(timeout_s=15.0):
                 self.writable = bool(onwm)
                 asyncio.get_running_loop().create_task(
                     self._background_snapshot())
@@ -354,11 +354,16 @@ class DbManager:
                     last_progress = flush
                     deadline = time.monotonic() + self.replication_timeout_s
                 if time.monotonic() > deadline:
+                    # ref :2452-2460: the shard stays read-only on a
+                    # catch-up timeout — but the gate must stay ARMED so
+                    # writes open the moment the standby finally catches
+                    # up (e.g. after it finishes a full restore), not
+                    # only on the next topology change
                     self.log.error(
                         "standby did not catch up within the replication "
-                        "timeout; shard stays read-only",
+                        "timeout; shard stays read-only (still watching)",
                         standby=standby_name)
-                    return
+                    deadline = time.monotonic() + self.replication_timeout_s
                 await asyncio.sleep(self.repl_poll_s)
             self.engine.write_conf("primary", sync_name=standby_name,
                                    read_only=False)
@@ -468,10 +473,17 @@ class DbManager:
         self.engine.post_restore_fixup()
         self.engine.write_conf("standby", upstream_url=upstream["pgUrl"])
         await self._start_db()
-        verdict = await self._await_streaming(timeout_s=30.0)
+        # short wait only: blocking here stalls the FSM's whole event
+        # loop, and the upstream may itself be mid-failover — the
+        # background watchdog re-triggers the transition if the standby
+        # turns out diverged or the upstream stays unreachable
+        verdict = await self._await_streaming(timeout_s=3.0)
         if verdict != "streaming":
-            self.log.error("standby still not streaming after restore",
-                           verdict=verdict)
+            self.log.warn("standby not yet streaming after restore; "
+                          "watching in the background", verdict=verdict)
+            self._transition_task = \
+                asyncio.get_running_loop().create_task(
+                    self._standby_watch())
 
     # --------------------------------------------------------------- queries
     async def get_xlog_location(self) -> str:

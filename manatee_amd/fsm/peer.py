@@ -136,6 +136,16 @@ class ManateePeer:
             kind, payload = await self._events.get()
             try:
                 self._ingest(kind, payload)
+                # drain everything already queued BEFORE evaluating: a
+                # single evaluation can block for a long time (e.g. a
+                # snapshot restore), and acting per-event would replay
+                # stale decisions against a world that has since moved on
+                # (fresher activeChange/clusterState events sitting in
+                # the queue).  Coalescing makes every evaluation use the
+                # newest known state.
+                while not self._events.empty():
+                    k2, p2 = self._events.get_nowait()
+                    self._ingest(k2, p2)
                 if self._shutdown:
                     continue
                 await self._eval_cluster_state()
