@@ -569,3 +569,89 @@ def test_metrics_endpoint(cluster_dir):
         finally:
             c.stop()
     run(go())
+
+
+def test_kill_sync_before_repl_established(cluster_dir):
+    """MANATEE-212 regression (ref integ.test.js
+    MANATEE_212_killSyncBeforeRepl*): the sync is SIGKILLed while the
+    primary is still read-only waiting for it to catch up.  The primary
+    must replace it with the async and open writes — never deadlock on
+    the dead sync."""
+    async def go():
+        c = DevCluster(cluster_dir, n_peers=3, shard_name="1.m212")
+        try:
+            await c.start()
+            s = await c.wait_cluster(
+                lambda s: s.get("sync") and len(s.get("async", [])) == 1,
+                timeout_s=60, what="formation")
+            await c.wait_writable(timeout_s=60)
+            gen = s["generation"]
+            # force a sync swap: kill the current sync; as soon as the
+            # NEXT state names the async as sync (primary read-only,
+            # waiting for catch-up), kill THAT sync too before it can
+            # finish catching up
+            first_sync = c.peer_by_id(s["sync"]["id"])
+            first_sync.kill9()
+            s2 = await c.wait_cluster(
+                lambda s2: s2["generation"] > gen
+                and s2["sync"]["id"] == s["async"][0]["id"],
+                timeout_s=60, what="first sync replacement")
+            second_sync = c.peer_by_id(s2["sync"]["id"])
+            second_sync.kill9()
+            # no spare async: the shard must hold read-only (no unsafe
+            # takeover) until a peer returns…
+            await asyncio.sleep(3.0)
+            s3 = await c.cluster_state()
+            assert s3["primary"]["id"] == s["primary"]["id"]
+            # …then the first sync comes back and the shard heals
+            first_sync.start()
+            await c.wait_cluster(
+                lambda s4: s4.get("sync") is not None
+                and s4["sync"]["id"] == first_sync.id,
+                timeout_s=90, what="sync re-established")
+            await c.wait_writable(timeout_s=90)
+        finally:
+            c.stop()
+    run(go())
+
+
+def test_rapid_sequenced_kills_converge(cluster_dir):
+    """MANATEE-207-style no-wait kills: SIGKILL the primary, then
+    SIGKILL its successor the moment it takes over, without letting the
+    shard settle.  Once peers restart, the shard must converge with all
+    acknowledged writes intact."""
+    async def go():
+        c = DevCluster(cluster_dir, n_peers=3, shard_name="1.m207")
+        try:
+            await c.start()
+            s = await c.wait_cluster(
+                lambda s: s.get("sync") and len(s.get("async", [])) == 1,
+                timeout_s=60, what="formation")
+            prim = await c.wait_writable(timeout_s=60)
+            cli = prim.db_client()
+            for i in range(25):
+                await cli.put("m207-%d" % i, i)
+            await cli.close()
+            gen = s["generation"]
+            sync_peer = c.peer_by_id(s["sync"]["id"])
+
+            prim.kill9()
+            # the moment the sync declares itself primary, kill it too
+            await c.wait_cluster(
+                lambda s2: s2["generation"] > gen
+                and s2["primary"]["id"] == sync_peer.id,
+                timeout_s=60, what="first takeover")
+            sync_peer.kill9()
+
+            # restart both dead peers; the shard must converge writable
+            await asyncio.sleep(1.0)
+            prim.start()
+            sync_peer.start()
+            newp = await c.wait_writable(timeout_s=120)
+            cli = newp.db_client()
+            for i in range(25):
+                assert await cli.get("m207-%d" % i) == i
+            await cli.close()
+        finally:
+            c.stop()
+    run(go(), timeout=300)
