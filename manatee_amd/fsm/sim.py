@@ -231,6 +231,24 @@ class Simulator:
         self._log("add %s" % ip)
         await self.shard.add_peer(ip)
 
+    async def _op_expire_session(self) -> None:
+        """Server-side session expiry of a live peer that keeps running —
+        the ZK-blip case: the peer must rebuild its session, rejoin the
+        election, and re-evaluate (it may discover it was deposed)."""
+        live = [p for p in self.shard.peers.values()
+                if p.ip not in self.dead]
+        if not live:
+            return
+        victim = self.rng.choice(live)
+        cli = victim.zk._zk
+        if cli is None or not cli.session_id:
+            return
+        sess = self.shard.srv.sessions.get(cli.session_id)
+        if sess is None:
+            return
+        self._log("expire-session %s" % victim.ip)
+        self.shard.srv._expire_session(sess)
+
     async def _op_toggle_freeze(self) -> None:
         from ..adm import core as adm
         zk = await adm.create_zk_client(self.shard.srv.conn_str)
@@ -312,7 +330,8 @@ class Simulator:
     # --------------------------------------------------------------- runs
     async def run(self, steps: int = 30) -> dict:
         ops = [(self._op_kill, 4), (self._op_restart, 4),
-               (self._op_add, 1), (self._op_toggle_freeze, 1)]
+               (self._op_add, 1), (self._op_toggle_freeze, 1),
+               (self._op_expire_session, 2)]
         weighted = [op for op, w in ops for _ in range(w)]
         await self.shard.start(n_peers=self.n_peers)
         try:
