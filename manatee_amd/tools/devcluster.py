@@ -307,6 +307,17 @@ class DevCluster:
              "-j", os.path.join(self.base_dir, "zk-journal.jsonl")],
             env=env, stdout=logf, stderr=logf, start_new_session=True)
 
+    def kill_zk(self) -> None:
+        """SIGKILL the coordination server (full-ZK outage tier of the
+        reference's chaos plan, docs/test-plan.md)."""
+        if self.zk_proc is not None and self.zk_proc.poll() is None:
+            try:
+                os.killpg(self.zk_proc.pid, signal.SIGKILL)
+            except ProcessLookupError:
+                pass
+            self.zk_proc.wait()
+        self.zk_proc = None
+
     async def wait_zk(self, timeout_s: float = 15.0) -> None:
         deadline = time.monotonic() + timeout_s
         while True:

@@ -114,7 +114,7 @@ async def soak(minutes: float, seed: int, workdir: str) -> dict:
             s = await c.cluster_state()
             action = rng.choice(
                 ["kill_primary", "kill_sync", "kill_async",
-                 "kill_db_only", "pause_primary"])
+                 "kill_db_only", "pause_primary", "zk_outage"])
             stats["kills"][action] = stats["kills"].get(action, 0) + 1
             prim = c.peer_by_id(s["primary"]["id"])
             victim = None
@@ -132,6 +132,11 @@ async def soak(minutes: float, seed: int, workdir: str) -> dict:
                 prim.kill_db_only()
             elif action == "pause_primary":
                 prim.pause()
+            elif action == "zk_outage":
+                c.kill_zk()
+                await asyncio.sleep(rng.uniform(1.0, 4.0))
+                c.start_zk()
+                await c.wait_zk()
 
             try:
                 await c.wait_writable(timeout_s=120)
