@@ -123,3 +123,16 @@ def test_bench_contract_runs_and_reports():
     assert "INVALID" not in out["config"]
     # sanity: the run really did the steps inside the wall clock
     assert out["ms_per_step"] * out["steps"] / 1000.0 <= elapsed + 1
+
+
+def test_native_codec_is_loaded_and_used():
+    """On the dedicated box the in-tree C++ codec must be present (the
+    snapshot carries the built .so) and wal.py must be using it — the
+    replication path may not silently run the pure-Python fallback."""
+    from manatee_amd.db.waldb import wal as walmod
+    from manatee_amd.native import codec
+    assert codec is not None, "native codec extension missing on the box"
+    assert walmod._native is codec
+    payload = b"x" * 1000
+    frame = codec.encode_frame(payload)
+    assert list(walmod.parse_frames(frame)) == [(len(frame), payload)]
