@@ -99,6 +99,31 @@ class WaldbClient:
             raise WaldbError(resp.get("error", "get failed"))
         return resp.get("v") if resp.get("found") else None
 
+    async def delete(self, key: str, timeout_s: Optional[float] = None
+                     ) -> str:
+        resp = await self.query({"q": "del", "k": key},
+                                timeout_s=timeout_s)
+        if not resp.get("ok"):
+            raise WaldbError(resp.get("error", "del failed"))
+        return resp["lsn"]
+
+    async def batch(self, ops, timeout_s: Optional[float] = None) -> str:
+        """Atomic multi-op commit: ops = [{"op": "put"|"del", "k": ...,
+        "v": ...}, ...] — one WAL record, all-or-nothing."""
+        resp = await self.query({"q": "batch", "ops": list(ops)},
+                                timeout_s=timeout_s)
+        if not resp.get("ok"):
+            raise WaldbError(resp.get("error", "batch failed"))
+        return resp["lsn"]
+
+    async def scan(self, prefix: str = "", limit: int = 1000,
+                   after: Optional[str] = None) -> list:
+        resp = await self.query({"q": "scan", "prefix": prefix,
+                                 "limit": limit, "after": after})
+        if not resp.get("ok"):
+            raise WaldbError(resp.get("error", "scan failed"))
+        return resp["items"]
+
     async def count(self, prefix: Optional[str] = None,
                     timeout_s: Optional[float] = None) -> int:
         resp = await self.query({"q": "count", "prefix": prefix},

@@ -519,3 +519,92 @@ def test_follower_behind_recycled_wal_is_refused(tmp_path):
             stand.stop()
     finally:
         prim.stop()
+
+
+def test_del_batch_scan_replicated_and_crash_safe(tmp_path):
+    """del / atomic batch / prefix scan: WAL-logged, replicated to the
+    standby, atomic across kill -9 recovery."""
+    import shutil
+    prim = Node(tmp_path, "prim")
+    stby = Node(tmp_path, "bsync")
+    prim.init()
+    prim.write_conf(role="primary")
+    prim.start()
+    try:
+        async def seed():
+            c = prim.client()
+            for i in range(10):
+                await c.put("acct:%03d" % i, {"bal": i * 10})
+            await c.close()
+        run(seed())
+        shutil.copytree(prim.data_dir, stby.data_dir,
+                        ignore=shutil.ignore_patterns("waldb.pid",
+                                                      "waldb.conf"))
+        prim.write_conf(role="primary", sync_name="bsync")
+        prim.sighup()
+        stby.write_conf(role="standby",
+                        upstream="127.0.0.1:%d" % prim.port)
+        stby.start()
+
+        async def exercise():
+            c = prim.client()
+            sc = stby.client()
+
+            async def streaming():
+                st = await sc.status()
+                return st["upstream_status"] == "streaming"
+            await wait_async(streaming, what="standby streaming")
+
+            # delete
+            await c.delete("acct:000")
+            assert await c.get("acct:000") is None
+            assert await c.count(prefix="acct:") == 9
+
+            # atomic transfer: one batch record moves balance 001 -> 002
+            await c.batch([
+                {"op": "put", "k": "acct:001", "v": {"bal": 0}},
+                {"op": "put", "k": "acct:002", "v": {"bal": 30}},
+                {"op": "del", "k": "acct:003"},
+            ])
+            assert await c.get("acct:001") == {"bal": 0}
+            assert await c.get("acct:002") == {"bal": 30}
+            assert await c.get("acct:003") is None
+
+            # scan with pagination, and on the STANDBY
+            items = await c.scan(prefix="acct:", limit=3)
+            assert [i["k"] for i in items] == \
+                ["acct:001", "acct:002", "acct:004"]
+            more = await c.scan(prefix="acct:", after=items[-1]["k"])
+            assert more[0]["k"] == "acct:005"
+
+            async def replicated():
+                return await sc.get("acct:003") is None and \
+                    (await sc.get("acct:002")) == {"bal": 30}
+            await wait_async(replicated, what="batch on standby")
+            sitems = await sc.scan(prefix="acct:")
+            assert len(sitems) == 8
+            # standby rejects writes of every kind
+            with pytest.raises(WaldbError):
+                await sc.delete("acct:004")
+            with pytest.raises(WaldbError):
+                await sc.batch([{"op": "del", "k": "acct:004"}])
+            await c.close()
+            await sc.close()
+        run(exercise())
+
+        # crash recovery preserves the batch atomically
+        prim.kill9()
+        prim.write_conf(role="primary")   # no sync gate for the check
+        prim.start()
+
+        async def recovered():
+            c = prim.client()
+            assert await c.get("acct:001") == {"bal": 0}
+            assert await c.get("acct:002") == {"bal": 30}
+            assert await c.get("acct:003") is None
+            assert await c.count(prefix="acct:") == 8
+            await c.close()
+        run(recovered())
+    finally:
+        prim.stop()
+        stby.stop()
