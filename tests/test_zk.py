@@ -302,3 +302,23 @@ def test_journal_sequential_replay_and_compaction(tmp_path):
         await cli3.close()
         await srv3.stop()
     run(go())
+
+
+def test_ensemble_conn_string_skips_dead_servers():
+    """A multi-server connection string works when some addresses are
+    down: the client rotates to a live server (the ensemble behavior
+    the production configs rely on)."""
+    async def go():
+        srv = ZkServer()
+        await srv.start()
+        dead1 = "127.0.0.1:1"          # never listening
+        dead2 = "127.0.0.1:2"
+        conn = ",".join([dead1, dead2, srv.conn_str])
+        cli = ZkClient(conn, session_timeout_ms=4000)
+        await cli.connect(timeout_s=15)
+        await cli.mkdirp("/ens")
+        data, _ = await cli.get_data("/ens")
+        assert data == b""
+        await cli.close()
+        await srv.stop()
+    run(go())
