@@ -99,6 +99,46 @@ class WaldbClient:
             raise WaldbError(resp.get("error", "get failed"))
         return resp.get("v") if resp.get("found") else None
 
+    async def pipeline(self, reqs, timeout_s: Optional[float] = None
+                       ) -> list:
+        """Send many requests back-to-back and read the responses in
+        order — removes the per-request round trip (the wire protocol
+        answers strictly in order)."""
+        reqs = list(reqs)
+        async with self._lock:
+            await self._ensure()
+            try:
+                payload = b"".join(
+                    json.dumps(r).encode() + b"\n" for r in reqs)
+                self._writer.write(payload)
+                await self._writer.drain()
+                out = []
+                for _ in reqs:
+                    line = await asyncio.wait_for(
+                        self._reader.readline(),
+                        timeout_s if timeout_s is not None
+                        else self.query_timeout_s)
+                    if not line:
+                        raise WaldbError("connection closed by server")
+                    out.append(json.loads(line))
+                return out
+            except (ConnectionError, OSError, asyncio.TimeoutError,
+                    ValueError) as exc:
+                await self._teardown()
+                raise WaldbError("pipeline failed: %r" % (exc,)) from exc
+
+    async def put_many(self, items, timeout_s: Optional[float] = None
+                       ) -> int:
+        """Pipelined puts; items = iterable of (key, value).  Returns the
+        number acknowledged; raises on the first failed put."""
+        resps = await self.pipeline(
+            ({"q": "put", "k": k, "v": v} for k, v in items),
+            timeout_s=timeout_s)
+        for r in resps:
+            if not r.get("ok"):
+                raise WaldbError(r.get("error", "put failed"))
+        return len(resps)
+
     async def delete(self, key: str, timeout_s: Optional[float] = None
                      ) -> str:
         resp = await self.query({"q": "del", "k": key},

@@ -22,14 +22,21 @@ from ..db.waldb.client import WaldbClient
 from .devcluster import DevCluster
 
 
-async def writer_task(host, port, label, stop, counter):
+async def writer_task(host, port, label, stop, counter, pipeline=0):
     cli = WaldbClient(host, port)
     i = 0
     try:
         while not stop.is_set():
-            await cli.put("w-%s-%d" % (label, i), i)
-            counter[0] += 1
-            i += 1
+            if pipeline > 1:
+                n = await cli.put_many(
+                    (("w-%s-%d" % (label, i + j), i + j)
+                     for j in range(pipeline)))
+                counter[0] += n
+                i += pipeline
+            else:
+                await cli.put("w-%s-%d" % (label, i), i)
+                counter[0] += 1
+                i += 1
     finally:
         await cli.close()
 
@@ -46,7 +53,8 @@ async def reader_task(host, port, stop, counter):
         await cli.close()
 
 
-async def measure(c: DevCluster, n_writers: int, seconds: float) -> dict:
+async def measure(c: DevCluster, n_writers: int, seconds: float,
+                  pipeline: int = 0) -> dict:
     s = await c.cluster_state()
     prim = c.peer_by_id(s["primary"]["id"])
     sync = c.peer_by_id(s["sync"]["id"])
@@ -54,7 +62,8 @@ async def measure(c: DevCluster, n_writers: int, seconds: float) -> dict:
     wcount = [0]
     rcount = [0]
     tasks = [asyncio.ensure_future(
-        writer_task(prim.ip, prim.pg_port, str(w), stop, wcount))
+        writer_task(prim.ip, prim.pg_port, "%d.%d" % (pipeline, w),
+                    stop, wcount, pipeline=pipeline))
         for w in range(n_writers)]
     tasks += [asyncio.ensure_future(
         reader_task(sync.ip, sync.pg_port, stop, rcount))
@@ -64,7 +73,7 @@ async def measure(c: DevCluster, n_writers: int, seconds: float) -> dict:
     stop.set()
     await asyncio.gather(*tasks, return_exceptions=True)
     dt = time.monotonic() - t0
-    return {"writers": n_writers,
+    return {"writers": n_writers, "pipeline": pipeline,
             "acked_puts_per_s": round(wcount[0] / dt, 1),
             "standby_reads_per_s": round(rcount[0] / dt, 1)}
 
@@ -84,6 +93,11 @@ async def run(writers, seconds, workdir) -> dict:
             print("# writers=%d: %.0f acked puts/s, %.0f standby reads/s"
                   % (w, r["acked_puts_per_s"], r["standby_reads_per_s"]),
                   file=sys.stderr)
+        # pipelined bulk mode: batches of 50 per round trip
+        r = await measure(c, 4, seconds, pipeline=50)
+        results.append(r)
+        print("# writers=4 pipeline=50: %.0f acked puts/s"
+              % r["acked_puts_per_s"], file=sys.stderr)
         # durability sanity: everything acked must be present
         s = await c.cluster_state()
         cli = c.peer_by_id(s["primary"]["id"]).db_client()

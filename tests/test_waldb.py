@@ -608,3 +608,27 @@ def test_del_batch_scan_replicated_and_crash_safe(tmp_path):
     finally:
         prim.stop()
         stby.stop()
+
+
+def test_pipelined_puts(tmp_path):
+    n = Node(tmp_path, "pipe")
+    n.init()
+    n.write_conf(role="primary")
+    n.start()
+    try:
+        async def go():
+            c = n.client()
+            assert await c.put_many((("p%03d" % i, i)
+                                     for i in range(500))) == 500
+            assert await c.count(prefix="p") == 500
+            assert await c.get("p499") == 499
+            # a failed op in the pipeline raises, earlier ops committed
+            n.write_conf(role="primary", read_only=True)
+            n.sighup()
+            await asyncio.sleep(0.2)
+            with pytest.raises(WaldbError):
+                await c.put_many([("q1", 1)])
+            await c.close()
+        run(go())
+    finally:
+        n.stop()
