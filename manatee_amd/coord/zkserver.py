@@ -181,7 +181,14 @@ class ZkServer:
                 line = line.strip()
                 if not line:
                     continue
-                op = json.loads(line)
+                try:
+                    op = json.loads(line)
+                except ValueError:
+                    # torn tail from a dirty kill mid-append; nothing
+                    # after it can be valid either
+                    self.log.warn("journal line unparseable; stopping "
+                                  "replay", entries=n)
+                    break
                 try:
                     if op["op"] == "create":
                         # journaled paths are FINAL (already sequenced):

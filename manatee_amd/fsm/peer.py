@@ -72,6 +72,7 @@ class ManateePeer:
 
         self._zk_inited = False
         self._db_inited = False
+        self._db_setup = False
         self._db_online = False
         self._cluster_state: Optional[dict] = None
         self._actives: List[dict] = []
@@ -167,8 +168,9 @@ class ManateePeer:
             self._cluster_state = payload
         elif kind == "db-init":
             self._db_inited = True
+            self._db_setup = bool(payload.get("setup"))
             self._db_online = bool(payload.get("online"))
-            self.log.info("db init", setup=payload.get("setup"),
+            self.log.info("db init", setup=self._db_setup,
                           online=self._db_online)
         elif kind == "db-healthy":
             self._db_online = True
@@ -304,6 +306,24 @@ class ManateePeer:
             }
             if await self._write_state(new_state, "onwm cluster setup"):
                 await self._eval_cluster_state()
+            return
+
+        # SAFETY: a peer whose database already holds data must never
+        # auto-declare a new generation — if the coordination state is
+        # gone but the data is not, re-forming around arbitrary election
+        # order could elect a STALE peer as primary and destroy
+        # acknowledged writes when the up-to-date peers re-slave to it.
+        # This is the operator's `manatee-adm state-backfill` situation
+        # (ref stateBackfill lib/adm.js:1231-1312, which auto-freezes for
+        # the same reason).
+        if self._db_setup:
+            self._peer_state = "waiting (no cluster state, but database " \
+                "is initialized)"
+            self._warn_throttled(
+                "setup-initialized",
+                "database is initialized but no cluster state exists; "
+                "refusing to auto-form — run `manatee-adm state-backfill` "
+                "(or clear the data dir) to proceed")
             return
 
         actives = self._actives

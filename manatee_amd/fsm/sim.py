@@ -55,7 +55,9 @@ class MockDb:
         for cb in self._listeners.get(event, []):
             cb(*args)
 
-    def fire_init(self, setup=True, online=False):
+    def fire_init(self, setup=False, online=False):
+        # setup=False: a fresh database (the normal formation case);
+        # peers with already-initialized databases refuse to auto-form
         self.emit("init", {"setup": setup, "online": online})
 
     async def reconfigure(self, cfg):

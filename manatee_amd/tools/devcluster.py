@@ -128,15 +128,27 @@ class DevPeer:
     def db_pids(self) -> List[int]:
         """Candidate db pids: the manager-written db_child.pid (written at
         spawn time, so it can never lag the child) plus the db's own pid
-        file (covers children of sitters from earlier incarnations)."""
+        file (covers children of sitters from earlier incarnations).
+        Every pid is verified against /proc/<pid>/cmdline (must be a db
+        server running on THIS peer's data dir) — pid files go stale and
+        pids get recycled, and SIGKILLing a recycled pid would murder an
+        innocent process."""
         data = os.path.join(self.store_dir, "live", "data")
         pids = []
         for name in ("db_child.pid", "waldb.pid"):
             try:
                 with open(os.path.join(data, name)) as f:
-                    pids.append(int(f.read().split()[0]))
+                    pid = int(f.read().split()[0])
             except (OSError, ValueError, IndexError):
-                pass
+                continue
+            try:
+                with open("/proc/%d/cmdline" % pid, "rb") as f:
+                    cmdline = f.read().replace(b"\x00", b" ").decode(
+                        "utf-8", "replace")
+            except OSError:
+                continue        # no such process
+            if "waldb" in cmdline and data in cmdline:
+                pids.append(pid)
         return pids
 
     def db_pid(self) -> Optional[int]:

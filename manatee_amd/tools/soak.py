@@ -49,13 +49,17 @@ class SoakWriter:
                         continue
                     peer = self.cluster.peer_by_id(s["primary"]["id"])
                     cli = peer.db_client()
-                key = "soak-%d" % self.seq
-                await cli.put(key, self.seq, timeout_s=1.0)
-                self.window[key] = self.seq
-                self.acked_count += 1
-                if len(self.window) > WINDOW:
+                # pipelined burst: high sustained write pressure
+                base = self.seq
+                items = [("soak-%d" % (base + j), base + j)
+                         for j in range(25)]
+                await cli.put_many(items, timeout_s=2.0)
+                for k, v in items:
+                    self.window[k] = v
+                self.acked_count += len(items)
+                while len(self.window) > WINDOW:
                     self.window.pop(next(iter(self.window)))
-                self.seq += 1
+                self.seq += len(items)
             except Exception:
                 if cli is not None:
                     await cli.close()
@@ -74,23 +78,29 @@ class SoakWriter:
 
 
 async def verify(cluster: DevCluster, writer: SoakWriter) -> int:
-    s = await cluster.cluster_state()
-    cli = cluster.peer_by_id(s["primary"]["id"]).db_client()
-    lost = 0
-    try:
-        present = await cli.count(prefix="soak-", timeout_s=30.0)
-        if present < writer.acked_count:
-            lost += writer.acked_count - present
-        # spot-check a sample of the recent window (the exact count above
-        # already catches any missing key; this guards values)
-        items = list(writer.window.items())
-        sample = items[-200:] + items[:50]
-        for key, val in sample:
-            if await cli.get(key) != val:
-                lost += 1
-    finally:
-        await cli.close()
-    return lost
+    last_exc = None
+    for _attempt in range(3):
+        s = await cluster.cluster_state()
+        cli = cluster.peer_by_id(s["primary"]["id"]).db_client()
+        lost = 0
+        try:
+            present = await cli.count(prefix="soak-", timeout_s=30.0)
+            if present < writer.acked_count:
+                lost += writer.acked_count - present
+            # spot-check a sample of the recent window (the exact count
+            # above already catches any missing key; this guards values)
+            items = list(writer.window.items())
+            sample = items[-200:] + items[:50]
+            for key, val in sample:
+                if await cli.get(key) != val:
+                    lost += 1
+            return lost
+        except Exception as exc:     # mid-verify failover: retry fresh
+            last_exc = exc
+            await asyncio.sleep(1.0)
+        finally:
+            await cli.close()
+    raise RuntimeError("verification failed repeatedly: %r" % last_exc)
 
 
 async def soak(minutes: float, seed: int, workdir: str) -> dict:

@@ -443,3 +443,44 @@ def test_everyone_dies_cluster_resumes_on_return():
         finally:
             await shard.stop()
     run(go())
+
+
+def test_initialized_db_refuses_autoformation():
+    """SAFETY: peers whose databases already hold data must never
+    auto-declare generation 1 when cluster state is missing — arbitrary
+    election order could elect a stale peer as primary and acknowledged
+    writes would be destroyed when the others re-slave to it.  This is
+    the operator state-backfill situation."""
+    from tests.harness import Shard
+
+    async def go():
+        shard = Shard()
+        shard.srv = __import__("manatee_amd.coord.zkserver",
+                               fromlist=["ZkServer"]).ZkServer(
+            tick_ms=50, min_session_timeout_ms=300)
+        await shard.srv.start()
+        try:
+            for i in range(3):
+                p = await shard.add_peer("10.0.0.%d" % (i + 1))
+                # override the fresh-db default: data already exists
+                p.db.fire_init(setup=True)
+            await asyncio.sleep(2.0)
+            assert await shard.state() is None, \
+                "peers auto-formed despite initialized databases"
+            # the operator backfills; the cluster then proceeds
+            from manatee_amd.adm import core as adm
+            zk = await adm.create_zk_client(shard.srv.conn_str)
+            try:
+                state = await adm.state_backfill(
+                    zk, "/manatee/simshard")
+                assert state["generation"] == 0
+                assert state["freeze"], "backfill must auto-freeze"
+            finally:
+                await zk.close()
+            s = await shard.wait_state(
+                lambda s: s.get("primary") is not None,
+                what="state after backfill")
+            assert s["generation"] == 0
+        finally:
+            await shard.stop()
+    run(go())
