@@ -236,6 +236,25 @@ class Wal:
             n -= take
         return bytes(out)
 
+    def read_aligned(self, start: int, max_bytes: int = 1 << 20) -> bytes:
+        """Raw WAL bytes from ``start``, truncated to WHOLE records —
+        the only safe unit for replication (a receiver applies chunk by
+        chunk; a torn frame at a window boundary would be appended but
+        never applied).  Grows the window if a single record exceeds
+        ``max_bytes``; returns b"" at end-of-WAL."""
+        while True:
+            buf = self.read(start, max_bytes)
+            if not buf:
+                return b""
+            valid, _records = _scan(buf)
+            if valid > 0:
+                return buf[:valid]
+            if start + len(buf) >= self.end:
+                # a torn tail at the very end of WAL cannot happen for
+                # committed records; nothing streamable yet
+                return b""
+            max_bytes *= 2   # one record larger than the window
+
     def iter_records(self, start: int = 0
                      ) -> Iterator[Tuple[int, bytes]]:
         """Yield (commit_lsn, payload) from offset ``start`` (must be a

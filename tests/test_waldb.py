@@ -632,3 +632,65 @@ def test_pipelined_puts(tmp_path):
         run(go())
     finally:
         n.stop()
+
+
+def test_standby_catches_up_from_far_behind(tmp_path):
+    """Regression: the replication stream must be record-aligned.  A
+    standby joining multiple MiB behind used to receive 1 MiB windows
+    that cut records at the boundary — the bytes landed in its WAL (so
+    its acks looked caught-up) but were never APPLIED, and a later
+    promote served a stale kv.  The standby must apply everything."""
+    import shutil
+    prim = Node(tmp_path, "prim")
+    late = Node(tmp_path, "late")
+    prim.init()
+    prim.write_conf(role="primary")
+    prim.start()
+    try:
+        async def seed_small():
+            c = prim.client()
+            await c.put("seed", 0)
+            await c.close()
+        run(seed_small())
+        # snapshot the dataset EARLY — the standby will start from here
+        shutil.copytree(prim.data_dir, late.data_dir,
+                        ignore=shutil.ignore_patterns("waldb.pid",
+                                                      "db_child.pid",
+                                                      "waldb.conf"))
+
+        async def bulk():
+            c = prim.client()
+            payload = "x" * 120
+            for start in range(0, 30000, 500):
+                await c.put_many((("blk%05d" % i, payload)
+                                  for i in range(start, start + 500)))
+            st = await c.status()
+            await c.close()
+            return st["current_lsn"]
+        end_lsn = run(bulk())
+        # the backlog really is multiple read-windows deep
+        assert int(end_lsn.split("/")[1], 16) > 3 * (1 << 20)
+
+        late.write_conf(role="standby",
+                        upstream="127.0.0.1:%d" % prim.port)
+        late.start()
+
+        async def verify():
+            sc = late.client()
+
+            async def caught_up():
+                st = await sc.status()
+                return st["upstream_status"] == "streaming" and \
+                    st["current_lsn"] == end_lsn
+            await wait_async(caught_up, timeout=60,
+                             what="standby WAL catch-up")
+            # THE critical check: the kv must have APPLIED everything,
+            # not merely appended it to the WAL
+            assert await sc.count(prefix="blk") == 30000
+            assert await sc.get("blk29999") == "x" * 120
+            assert await sc.get("blk00000") == "x" * 120
+            await sc.close()
+        run(verify())
+    finally:
+        prim.stop()
+        late.stop()
