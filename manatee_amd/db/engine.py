@@ -147,11 +147,22 @@ class WaldbEngine(Engine):
                           waldb_server.PROMOTE_TRIGGER), "w").close()
 
     def post_restore_fixup(self) -> None:
-        for name in ("waldb.pid", "db_child.pid"):
+        # purge PEER-LOCAL runtime files that rode in with the snapshot:
+        # the SOURCE peer's conf (its port/role!), pid files and log.
+        # The conf is regenerated for THIS peer right after; a foreign
+        # conf left in place could make a db adopt another peer's
+        # identity (role=primary on the wrong port)
+        for name in ("waldb.pid", "db_child.pid", waldb_server.CONF_NAME,
+                     waldb_server.PROMOTE_TRIGGER):
             try:
                 os.unlink(os.path.join(self.data_dir, name))
             except FileNotFoundError:
                 pass
+        try:
+            os.unlink(os.path.join(os.path.dirname(self.data_dir),
+                                   "waldb.log"))
+        except FileNotFoundError:
+            pass
 
     def spawn_argv(self) -> List[str]:
         return [sys.executable, "-m", "manatee_amd.db.waldb.server",

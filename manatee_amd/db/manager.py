@@ -136,25 +136,48 @@ class DbManager:
                 self._emit("unhealthy")
 
     # ------------------------------------------------------- process control
+    def _pid_paths(self) -> List[str]:
+        paths = []
+        try:
+            paths.append(os.path.join(self.engine.data_dir,
+                                      "db_child.pid"))
+        except AttributeError:
+            pass
+        try:
+            paths.append(os.path.join(
+                os.path.dirname(self.store.mountpoint()), "db_child.pid"))
+        except Exception:
+            pass
+        return paths
+
     async def _start_db(self) -> None:
         if self._proc is not None:
             return
         self._expect_exit = False
         argv = self.engine.spawn_argv()
         self.log.info("starting database", argv=argv)
+        def _child_init():
+            # the db installs its own SIGHUP handler once its event loop
+            # is up; until then a reload signal must be ignored, not
+            # fatal (default SIGHUP disposition kills the interpreter)
+            signal.signal(signal.SIGHUP, signal.SIG_IGN)
+
         self._proc = await asyncio.create_subprocess_exec(
             *argv, stdout=asyncio.subprocess.DEVNULL,
             stderr=asyncio.subprocess.DEVNULL,
-            start_new_session=True)
+            start_new_session=True, preexec_fn=_child_init)
         # record the child pid IMMEDIATELY: the db writes its own pid file
         # only once it is up, and anything that needs to SIGKILL the whole
-        # peer (tests, operators) must not race that window
-        try:
-            with open(os.path.join(self.engine.data_dir, "db_child.pid"),
-                      "w") as f:
-                f.write(str(self._proc.pid))
-        except (OSError, AttributeError):
-            pass
+        # peer (tests, operators) must not race that window.  Written both
+        # inside the data dir AND outside the dataset (the data dir is
+        # REPLACED by restores, which would orphan the record and let an
+        # old child survive a kill, squatting on the port)
+        for pid_path in self._pid_paths():
+            try:
+                with open(pid_path, "w") as f:
+                    f.write(str(self._proc.pid))
+            except OSError:
+                pass
         self._proc_monitor = asyncio.get_running_loop().create_task(
             self._monitor_proc(self._proc))
         # poll until the db answers (ref _start 1 Hz poll :1760-1794;
@@ -164,7 +187,29 @@ class DbManager:
             if self._proc is None or self._proc.returncode is not None:
                 raise RuntimeError("database exited during startup")
             if await self.engine.ping(timeout_s=1.0):
-                break
+                # verify the responder IS our child: an older incarnation
+                # that escaped a kill can still hold the port, answer
+                # pings, and leave our fresh child dead on a bind failure
+                try:
+                    status = await self.engine.status()
+                except Exception:
+                    await asyncio.sleep(0.05)
+                    continue
+                serving = status.get("pid")
+                if serving in (None, self._proc.pid):
+                    break
+                self.log.error("a previous database incarnation still "
+                               "holds the port; killing it",
+                               old_pid=serving, new_pid=self._proc.pid)
+                try:
+                    os.kill(serving, signal.SIGKILL)
+                except (ProcessLookupError, PermissionError):
+                    pass
+                # our own child likely died on the bind conflict; respawn
+                if self._proc.returncode is not None:
+                    self._proc = None
+                    self._expect_exit = True
+                    return await self._start_db()
             if time.monotonic() > deadline:
                 raise RuntimeError("database did not become ready in %ss"
                                    % self.ops_timeout_s)
@@ -338,14 +383,24 @@ class DbManager:
         (ref _waitForStandby :1037-1105 + _checkRepl :2390-2475)."""
         standby_name = peer_id_from_urls(downstream["pgUrl"],
                                          downstream["backupUrl"])
+        my_proc = self._proc      # the db incarnation this gate belongs to
         deadline = time.monotonic() + self.replication_timeout_s
         last_progress: Optional[str] = None
+
+        def stale() -> bool:
+            # a cancelled-but-racing gate from a previous transition must
+            # never reload/flip a NEWER db incarnation (a SIGHUP during
+            # interpreter startup would kill it outright)
+            return (self._transition_task is not asyncio.current_task()
+                    or self._proc is not my_proc)
         try:
             while True:
                 try:
                     repl = await self.engine.check_repl(standby_name)
                 except Exception:
                     repl = {"connected": False, "caught_up": False}
+                if stale():
+                    return
                 if repl.get("caught_up"):
                     break
                 # forward progress resets the timeout (ref :2452-2460)
@@ -365,6 +420,8 @@ class DbManager:
                         standby=standby_name)
                     deadline = time.monotonic() + self.replication_timeout_s
                 await asyncio.sleep(self.repl_poll_s)
+            if stale():
+                return
             self.engine.write_conf("primary", sync_name=standby_name,
                                    read_only=False)
             self._reload_db()

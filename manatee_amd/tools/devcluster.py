@@ -135,9 +135,13 @@ class DevPeer:
         innocent process."""
         data = os.path.join(self.store_dir, "live", "data")
         pids = []
-        for name in ("db_child.pid", "waldb.pid"):
+        candidates = [os.path.join(data, "db_child.pid"),
+                      os.path.join(data, "waldb.pid"),
+                      # survives dataset replacement by restores
+                      os.path.join(self.store_dir, "db_child.pid")]
+        for path in candidates:
             try:
-                with open(os.path.join(data, name)) as f:
+                with open(path) as f:
                     pid = int(f.read().split()[0])
             except (OSError, ValueError, IndexError):
                 continue
@@ -147,7 +151,8 @@ class DevPeer:
                         "utf-8", "replace")
             except OSError:
                 continue        # no such process
-            if "waldb" in cmdline and data in cmdline:
+            if "waldb" in cmdline and data in cmdline \
+                    and pid not in pids:
                 pids.append(pid)
         return pids
 
