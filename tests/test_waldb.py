@@ -694,3 +694,96 @@ def test_standby_catches_up_from_far_behind(tmp_path):
     finally:
         prim.stop()
         late.stop()
+
+
+def test_sighup_during_boot_is_survivable(tmp_path):
+    """A reload signal delivered before the server installs its handler
+    must not kill it (children spawn with SIGHUP ignored; the conf
+    mtime watcher covers any missed reload)."""
+    import subprocess as sp
+    n = Node(tmp_path, "hup")
+    n.init()
+    n.write_conf(role="primary")
+    env = dict(os.environ, PYTHONPATH=REPO)
+
+    def child_init():
+        signal.signal(signal.SIGHUP, signal.SIG_IGN)
+    proc = sp.Popen(
+        [sys.executable, "-m", "manatee_amd.db.waldb.server",
+         "-D", n.data_dir],
+        env=env, stdout=sp.DEVNULL, stderr=sp.DEVNULL,
+        preexec_fn=child_init)
+    try:
+        # storm of reload signals through the boot window
+        for _ in range(50):
+            proc.send_signal(signal.SIGHUP)
+            time.sleep(0.01)
+        deadline = time.time() + 15
+        pid_file = os.path.join(n.data_dir, "waldb.pid")
+        while not os.path.exists(pid_file) or \
+                os.stat(pid_file).st_size == 0:
+            assert proc.poll() is None, \
+                "server died from a boot-window SIGHUP"
+            assert time.time() < deadline
+            time.sleep(0.02)
+        # and the mtime watcher still applies conf changes without signal
+        n.write_conf(role="primary", read_only=True)
+
+        async def becomes_ro():
+            c = n.client()
+
+            async def ro():
+                st = await c.status()
+                return st["read_only"]
+            await wait_async(ro, timeout=10, what="mtime-watched reload")
+            await c.close()
+        run(becomes_ro())
+    finally:
+        proc.kill()
+        proc.wait()
+
+
+def test_foreign_conf_reload_refused(tmp_path):
+    """A conf naming a DIFFERENT port (e.g. another peer's conf that
+    rode in with a dataset restore) must never be adopted by a running
+    server."""
+    n = Node(tmp_path, "foreign")
+    n.init()
+    n.write_conf(role="primary")
+    n.start()
+    try:
+        async def go():
+            c = n.client()
+            await c.put("mine", 1)
+            # drop in a foreign conf: different port, standby role
+            confparser.write(os.path.join(n.data_dir, "waldb.conf"), {
+                "role": "standby", "listen_ip": "127.0.0.1",
+                "port": str(n.port + 1), "name": "someone-else",
+                "primary_conninfo": "'127.0.0.1:1'",
+            })
+            n.sighup()
+            await asyncio.sleep(0.5)
+            st = await c.status()
+            assert st["role"] == "primary", "adopted a foreign conf!"
+            await c.put("still-mine", 2)
+            await c.close()
+        run(go())
+    finally:
+        n.stop()
+
+
+def test_post_restore_fixup_purges_peer_local_files(tmp_path):
+    from manatee_amd.db.engine import WaldbEngine
+    data = str(tmp_path / "store" / "live" / "data")
+    os.makedirs(data)
+    for name in ("waldb.pid", "db_child.pid", "waldb.conf", "promote"):
+        open(os.path.join(data, name), "w").write("junk")
+    open(os.path.join(tmp_path / "store" / "live", "waldb.log"),
+         "w").write("foreign log")
+    open(os.path.join(data, "waldb_ident.json"), "w").write("{}")
+    eng = WaldbEngine(data, "127.0.0.1", 1, "p")
+    eng.post_restore_fixup()
+    left = sorted(os.listdir(data))
+    assert left == ["waldb_ident.json"], left
+    assert not os.path.exists(
+        os.path.join(tmp_path / "store" / "live", "waldb.log"))
