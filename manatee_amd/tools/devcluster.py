@@ -61,6 +61,7 @@ class DevPeer:
         self.store_dir = os.path.join(self.dir, "store")
         self.sitter_proc: Optional[subprocess.Popen] = None
         self.backup_proc: Optional[subprocess.Popen] = None
+        self.snap_proc: Optional[subprocess.Popen] = None
 
     # ------------------------------------------------------------- configs
     def sitter_config(self) -> dict:
@@ -98,12 +99,25 @@ class DevPeer:
             },
         }
 
+    def snapshotter_config(self) -> dict:
+        c = self.cluster
+        return {
+            "storageCfg": {"provider": "dir",
+                           "mountpoint": os.path.join(self.store_dir)},
+            "pollInterval": c.snapshot_interval_ms,
+            "snapshotNumber": c.snapshot_number,
+            "statusUrl": "http://%s:%d/ping" % (self.ip,
+                                                self.status_port),
+        }
+
     def write_configs(self) -> None:
         os.makedirs(self.dir, exist_ok=True)
         with open(os.path.join(self.dir, "sitter.json"), "w") as f:
             json.dump(self.sitter_config(), f, indent=2)
         with open(os.path.join(self.dir, "backupserver.json"), "w") as f:
             json.dump(self.backupserver_config(), f, indent=2)
+        with open(os.path.join(self.dir, "snapshotter.json"), "w") as f:
+            json.dump(self.snapshotter_config(), f, indent=2)
 
     # ------------------------------------------------------------- control
     def _spawn(self, module: str, config: str, logname: str
@@ -124,6 +138,10 @@ class DevPeer:
                                        "backupserver.json", "backupserver.log")
         self.sitter_proc = self._spawn("manatee_amd.daemons.sitter",
                                        "sitter.json", "sitter.log")
+        if self.cluster.run_snapshotter:
+            self.snap_proc = self._spawn("manatee_amd.daemons.snapshotter",
+                                         "snapshotter.json",
+                                         "snapshotter.log")
 
     def db_pids(self) -> List[int]:
         """Candidate db pids: the manager-written db_child.pid (written at
@@ -164,7 +182,7 @@ class DevPeer:
         """SIGKILL the whole peer: sitter process group + db child +
         backupserver — the integ-test failure mode
         (ref test/integ.test.js primaryDeath et al)."""
-        for proc in (self.sitter_proc, self.backup_proc):
+        for proc in (self.sitter_proc, self.backup_proc, self.snap_proc):
             if proc is not None and proc.poll() is None:
                 try:
                     os.killpg(proc.pid, signal.SIGKILL)
@@ -184,13 +202,15 @@ class DevPeer:
                     pass
         self.sitter_proc = None
         self.backup_proc = None
+        self.snap_proc = None
 
     def pause(self) -> None:
         """SIGSTOP the whole peer (sitter pg + db pg) — the network-
         partition analogue on one host: processes stay alive but stop
         responding, the ZK session expires, and on resume() the peer
         discovers the cluster moved on without it."""
-        pids = [p.pid for p in (self.sitter_proc, self.backup_proc)
+        pids = [p.pid for p in (self.sitter_proc, self.backup_proc,
+                                self.snap_proc)
                 if p is not None and p.poll() is None]
         pids += self.db_pids()
         for pid in pids:
@@ -203,7 +223,8 @@ class DevPeer:
                     pass
 
     def resume(self) -> None:
-        pids = [p.pid for p in (self.sitter_proc, self.backup_proc)
+        pids = [p.pid for p in (self.sitter_proc, self.backup_proc,
+                                self.snap_proc)
                 if p is not None and p.poll() is None]
         pids += self.db_pids()
         for pid in pids:
@@ -255,7 +276,10 @@ class DevCluster:
                  replication_timeout_ms: int = 30000,
                  tick_interval_ms: int = 250,
                  singleton: bool = False,
-                 base_port: Optional[int] = None):
+                 base_port: Optional[int] = None,
+                 run_snapshotter: bool = True,
+                 snapshot_interval_ms: int = 30000,
+                 snapshot_number: int = 5):
         self.base_dir = os.path.abspath(base_dir)
         self.ip = ip
         self.engine = engine
@@ -267,6 +291,9 @@ class DevCluster:
         self.replication_timeout_ms = replication_timeout_ms
         self.tick_interval_ms = tick_interval_ms
         self.singleton = singleton
+        self.run_snapshotter = run_snapshotter
+        self.snapshot_interval_ms = snapshot_interval_ms
+        self.snapshot_number = snapshot_number
         self.zk_port = 0  # assigned below the ephemeral range in __init__
         self.zk_conn_str = ""
         self.zk_proc: Optional[subprocess.Popen] = None
