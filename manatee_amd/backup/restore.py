@@ -18,7 +18,7 @@ from __future__ import annotations
 
 import asyncio
 import time
-from typing import Optional
+from typing import List, Optional
 
 from ..common.httpd import http_request
 from ..common.logging import Logger, null_logger
@@ -75,13 +75,18 @@ class RestoreClient:
             recv_done: asyncio.Future = asyncio.get_running_loop() \
                 .create_future()
             conn_seen = asyncio.Event()
+            conn_task: List[Optional[asyncio.Task]] = [None]
+            abandoned = [False]
 
             async def on_conn(reader: asyncio.StreamReader,
                               writer: asyncio.StreamWriter):
-                if conn_seen.is_set():
+                if conn_seen.is_set() or abandoned[0]:
+                    # a late sender callback for a restore that has moved
+                    # on must NEVER be allowed to touch the dataset
                     writer.close()
                     return
                 conn_seen.set()
+                conn_task[0] = asyncio.current_task()
 
                 async def chunks():
                     while True:
@@ -136,7 +141,19 @@ class RestoreClient:
                 # sender is done; wait for our receive side to finish
                 await asyncio.wait_for(recv_done, 60)
             finally:
+                # no receive may outlive this restore: a late-arriving or
+                # still-running stream task writing into the dataset
+                # AFTER we return would clobber whatever replaced it
+                # (a newer restore, a regenerated conf, a running db)
+                abandoned[0] = True
                 server.close()
+                task = conn_task[0]
+                if task is not None and not task.done():
+                    task.cancel()
+                    try:
+                        await task
+                    except (asyncio.CancelledError, Exception):
+                        pass
                 await server.wait_closed()
 
             # initial snapshot of the restored dataset

@@ -99,9 +99,15 @@ class DirStore(SnapshotStore):
         return gen()
 
     async def recv(self, chunks: AsyncIterator[bytes]) -> None:
-        await self.ensure()
+        """Receive into a STAGING directory and swap it into place only
+        once the stream completed — a partial or late-cancelled stream
+        must never become visible in the live dataset."""
+        os.makedirs(self.base, exist_ok=True)
+        staging = os.path.join(self.base,
+                               ".recv-%d-%d" % (os.getpid(), time.time_ns()))
+        os.makedirs(staging)
         proc = await asyncio.create_subprocess_exec(
-            "tar", "-xf", "-", "-C", self.live,
+            "tar", "-xf", "-", "-C", staging,
             stdin=asyncio.subprocess.PIPE,
             stdout=asyncio.subprocess.DEVNULL,
             stderr=asyncio.subprocess.PIPE,
@@ -115,10 +121,17 @@ class DirStore(SnapshotStore):
             if rc != 0:
                 err = (await proc.stderr.read()).decode("utf-8", "replace")
                 raise procutil.ExecError(["tar", "-xf"], rc, "", err)
+            # atomic-ish swap: retire any current live dir, then rename
+            if os.path.isdir(self.live):
+                old = self.live + ".replaced-%d" % time.time_ns()
+                os.rename(self.live, old)
+                shutil.rmtree(old, ignore_errors=True)
+            os.rename(staging, self.live)
         except BaseException:
             if proc.returncode is None:
                 proc.kill()
                 await proc.wait()
+            shutil.rmtree(staging, ignore_errors=True)
             raise
 
     # ------------------------------------------------------------ isolation
