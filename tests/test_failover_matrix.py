@@ -476,23 +476,32 @@ def test_5peer_cascading_chain_and_middle_death(cluster_dir):
                 timeout_s=90, what="5-peer formation")
             prim = await c.wait_writable(timeout_s=90)
 
-            # verify the replication edges: each peer's downstream row
-            # names its chain successor
-            chain = [s["primary"], s["sync"]] + s["async"]
-            for up, down in zip(chain[:-1], chain[1:]):
-                peer = c.peer_by_id(up["id"])
-                cli = peer.db_client()
-                deadline = time.monotonic() + 30
-                ok = False
-                while time.monotonic() < deadline and not ok:
-                    st = await cli.status()
-                    ok = any(r["application_name"] == down["id"]
-                             and r["state"] == "streaming"
-                             for r in st.get("replication", []))
-                    if not ok:
-                        await asyncio.sleep(0.2)
-                await cli.close()
-                assert ok, "no stream %s -> %s" % (up["id"], down["id"])
+            # verify the replication edges against the CURRENT state
+            # (async ordering can be revised while peers settle): each
+            # peer must stream to its chain successor
+            deadline = time.monotonic() + 60
+            missing = None
+            while time.monotonic() < deadline:
+                s = await c.cluster_state()
+                chain = [s["primary"], s["sync"]] + s["async"]
+                missing = None
+                for up, down in zip(chain[:-1], chain[1:]):
+                    cli = c.peer_by_id(up["id"]).db_client()
+                    try:
+                        st = await cli.status()
+                    except Exception:
+                        st = {}
+                    finally:
+                        await cli.close()
+                    if not any(r["application_name"] == down["id"]
+                               and r["state"] == "streaming"
+                               for r in st.get("replication", [])):
+                        missing = "%s -> %s" % (up["id"], down["id"])
+                        break
+                if missing is None:
+                    break
+                await asyncio.sleep(0.3)
+            assert missing is None, "no stream " + missing
 
             # a write reaches the chain tail
             cli = prim.db_client()
