@@ -133,8 +133,12 @@ class ManateeClient:
                 await self._zk.connect(timeout_s=10.0)
                 backoff = 0.1
                 while not self._closing:
-                    await self._read_and_watch()
+                    # Clear BEFORE re-arming the watch: a notification
+                    # dispatched between watch registration and wait() must
+                    # survive until wait(), or the consumed one-shot watch is
+                    # never re-registered and we'd serve stale topology.
                     self._poke.clear()
+                    await self._read_and_watch()
                     await self._poke.wait()
             except asyncio.CancelledError:
                 return

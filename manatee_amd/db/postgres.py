@@ -45,12 +45,17 @@ DATA_CONF_NAME = "manatee-config.json"
 PROMOTE_TRIGGER = "promote"
 
 # fixed template values (ref etc/postgresql.conf; SURVEY.md §6 durability
-# floor): hot-standby WAL, remote_write sync commit
+# floor): hot-standby WAL, remote_write sync commit.  The reference turns
+# full_page_writes OFF because it assumes ZFS (copy-on-write — no torn
+# pages); on a plain filesystem (DirStore) that would risk unrecoverable
+# torn-page corruption after power loss, so the safe value is the default
+# here and write_conf() only relaxes it when the configured store is
+# copy-on-write (cfg key ``storeIsCow``, set by the sitter for ZfsStore).
 BASE_CONF = {
     "wal_level": "hot_standby",
     "hot_standby": "on",
     "synchronous_commit": "remote_write",
-    "full_page_writes": "off",
+    "full_page_writes": "on",
     "max_wal_senders": "10",
     "wal_keep_segments": "64",
 }
@@ -153,6 +158,7 @@ class PostgresEngine(Engine):
         self.tunables_file = cfg.get("tunablesFile")
         self.db_user = cfg.get("dbUser", "postgres")
         self.connect_timeout_s = cfg.get("pgConnectTimeout", 60)
+        self.store_is_cow = bool(cfg.get("storeIsCow", False))
         self.dataset_dir = os.path.dirname(os.path.abspath(data_dir))
         self.data_conf = os.path.join(self.dataset_dir, DATA_CONF_NAME)
         # resolved by resolve_versioned_paths():
@@ -278,6 +284,8 @@ class PostgresEngine(Engine):
         from the template — never edit in place (ref :2282-2336)."""
         self.resolve_versioned_paths()
         conf = self._template_conf()
+        if self.store_is_cow:
+            conf["full_page_writes"] = "off"
         conf.update(self.tunables)
         conf["listen_addresses"] = "'%s'" % self.ip
         conf["port"] = str(self.port)

@@ -56,14 +56,18 @@ def peer_identity(cfg: dict) -> dict:
 
 
 def build_engine(mgr_cfg: dict, ip: str, pg_port: int, peer_name: str,
-                 data_dir: str, log: Logger):
+                 data_dir: str, log: Logger, store=None):
     kind = mgr_cfg.get("engine", "waldb")
     if kind == "waldb":
         return WaldbEngine(data_dir, ip, pg_port, peer_name, log=log)
     if kind == "postgres":
         from .db.postgres import PostgresEngine
+        from .storage.zfsstore import ZfsStore
+        cfg = dict(mgr_cfg)
+        # full_page_writes may only be relaxed on a copy-on-write store
+        cfg.setdefault("storeIsCow", isinstance(store, ZfsStore))
         return PostgresEngine(data_dir, ip, pg_port, peer_name,
-                              cfg=mgr_cfg, log=log)
+                              cfg=cfg, log=log)
     raise ValueError("unknown engine %r" % kind)
 
 
@@ -78,7 +82,8 @@ class Shard:
         data_dir = mgr_cfg.get("dataDir") or \
             os.path.join(self.store.mountpoint(), "data")
         self.engine = build_engine(mgr_cfg, cfg["ip"], cfg["postgresPort"],
-                                   self.ident["id"], data_dir, self.log)
+                                   self.ident["id"], data_dir, self.log,
+                                   store=self.store)
         self.db_manager = DbManager(
             engine=self.engine, store=self.store, ip=cfg["ip"],
             health_interval_s=mgr_cfg.get("healthChkInterval", 1000) / 1000.0,
