@@ -63,12 +63,20 @@ class Writer:
         self.seq = 0
         self.task = None
         self.stop_flag = False
+        self.pause_flag = False
+        self._paused = asyncio.Event()
         self.last_ack_time = None
 
     async def _run(self):
         cli = None
         peer = None
         while not self.stop_flag:
+            if self.pause_flag:
+                # quiescent point: no put is in flight while paused, so
+                # acked_count is exact and frozen for the loss check
+                self._paused.set()
+                await asyncio.sleep(0.01)
+                continue
             try:
                 if cli is None:
                     s = await self.cluster.cluster_state()
@@ -97,11 +105,98 @@ class Writer:
         self.stop_flag = False
         self.task = asyncio.get_running_loop().create_task(self._run())
 
+    async def pause(self):
+        """Freeze the writer at an iteration boundary (no in-flight put).
+        While paused no new acks can arrive, so a server-side count
+        compared against ``acked_count`` is exact — post-failover acks
+        cannot pad over a lost write."""
+        self._paused.clear()
+        self.pause_flag = True
+        await self._paused.wait()
+
+    def resume(self):
+        self.pause_flag = False
+
     async def stop(self):
         self.stop_flag = True
+        self.pause_flag = False
         if self.task is not None:
             await self.task
             self.task = None
+
+
+# test seam: called with (cluster, state) after failover completes and
+# BEFORE the loss verification — the injected-loss regression test uses it
+# to delete an acknowledged key and prove the check catches it
+_pre_verify_hook = None
+
+
+async def verify_no_loss(cluster: DevCluster, state: dict,
+                         writer: Writer) -> dict:
+    """Exact zero-acknowledged-write-loss check.
+
+    The writer is FROZEN first (no in-flight put, no new acks), so the
+    acked set is exactly ``bench-0 .. bench-(seq-1)`` — the writer only
+    advances ``seq`` after an ack and retries the same key otherwise.
+    A server-side count of the prefix therefore detects ANY lost key;
+    the rolling window (the writes nearest the kill) is additionally
+    read back value-by-value.  The new SYNC must converge to hold every
+    acked write as well (it replicates from the new primary)."""
+    await writer.pause()
+    try:
+        acked_total = writer.acked_count
+        window = dict(writer.acked)
+        assert acked_total == writer.seq, \
+            "writer invariant broken: acked_count != seq while paused"
+
+        # at most ONE committed-but-unacked key can exist (the writer has a
+        # single in-flight put and retries the same seq until acked): it is
+        # exactly bench-<seq>.  Probe it and exclude it from the count so it
+        # cannot pad over a lost acked key.
+        probe_key = "bench-%d" % acked_total
+
+        newp = cluster.peer_by_id(state["primary"]["id"])
+        cli = newp.db_client()
+        lost = 0
+        try:
+            present = await cli.count(prefix="bench-")
+            if await cli.get(probe_key) is not None:
+                present -= 1
+            if present < acked_total:
+                lost += acked_total - present
+            for key, val in window.items():
+                got = await cli.get(key)
+                if got != val:
+                    lost += 1
+        finally:
+            await cli.close()
+
+        # the sync must CONVERGE to contain every acked write (it may be
+        # the old async, still replaying) — poll with a deadline
+        sync_lost = 0
+        if state.get("sync"):
+            syncp = cluster.peer_by_id(state["sync"]["id"])
+            scli = syncp.db_client()
+            try:
+                deadline = time.monotonic() + 15.0
+                sync_present = 0
+                while time.monotonic() < deadline:
+                    try:
+                        sync_present = await scli.count(prefix="bench-")
+                        if await scli.get(probe_key) is not None:
+                            sync_present -= 1
+                    except Exception:
+                        sync_present = 0
+                    if sync_present >= acked_total:
+                        break
+                    await asyncio.sleep(0.05)
+                if sync_present < acked_total:
+                    sync_lost = acked_total - sync_present
+            finally:
+                await scli.close()
+        return {"lost": lost + sync_lost, "checked": acked_total}
+    finally:
+        writer.resume()
 
 
 async def one_failover(cluster: DevCluster, writer: Writer) -> dict:
@@ -109,8 +204,6 @@ async def one_failover(cluster: DevCluster, writer: Writer) -> dict:
     s = await cluster.cluster_state()
     prim = cluster.peer_by_id(s["primary"]["id"])
     old_gen = s["generation"]
-    acked_before = dict(writer.acked)          # recent-window sample
-    acked_count_before = writer.acked_count    # exact total
 
     t_kill = time.monotonic()
     prim.kill9()
@@ -135,28 +228,17 @@ async def one_failover(cluster: DevCluster, writer: Writer) -> dict:
     if t_writable is None:
         raise RuntimeError("failover did not complete within 120 s")
 
-    # zero acknowledged-write-loss check against the new primary: the
-    # exact count of unique acked keys must all be present, and the
-    # recent window (the writes nearest the kill) must read back intact
     s2 = await cluster.cluster_state()
-    newp = cluster.peer_by_id(s2["primary"]["id"])
-    cli = newp.db_client()
-    lost = 0
-    present = await cli.count(prefix="bench-")
-    if present < acked_count_before:
-        lost += acked_count_before - present
-    for key, val in acked_before.items():
-        got = await cli.get(key)
-        if got != val:
-            lost += 1
-    await cli.close()
+    if _pre_verify_hook is not None:
+        await _pre_verify_hook(cluster, s2)
+    v = await verify_no_loss(cluster, s2, writer)
 
     # heal: rebuild the deposed ex-primary so the next step starts from a
     # full primary/sync/async shard
     await cluster.rebuild_peer(prim)
     await cluster.wait_writable(timeout_s=60)
-    return {"failover_s": t_writable - t_kill, "lost_acked_writes": lost,
-            "acked_checked": acked_count_before}
+    return {"failover_s": t_writable - t_kill,
+            "lost_acked_writes": v["lost"], "acked_checked": v["checked"]}
 
 
 async def run_rank(rank: int, steps: int, warmup: int, base_dir: str
