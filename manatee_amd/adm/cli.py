@@ -35,7 +35,62 @@ RESTORE_ATTEMPTS = 5          # ref lib/adm.js:71
 
 
 class UsageError(Exception):
-    pass
+    """CLI usage error: printed as ``manatee-adm: <msg>`` followed by the
+    subcommand's help text, exit 2 (the reference's cmdln usage()
+    behavior, which its golden tests capture byte-for-byte)."""
+
+    def __init__(self, msg: str, help_text: Optional[str] = None):
+        super().__init__(msg)
+        self.help_text = help_text
+
+
+# Byte-compatible help texts for the committed subcommands (rendered
+# exactly as the reference's cmdln/dashdash does — captured in
+# /root/reference/test/tst.manateeAdm.js.out and tst.manateeAdmUsage.js).
+_HELP_OPTS = {
+    "help": "    -h, --help                          Show this help.",
+    "columns": ("    -o COLNAME[,...], --columns=COLNAME[,...]\n"
+                "                                        Columns to print."),
+    "omitHeader":
+        "    -H, --omitHeader                    Omit header row from output.",
+    "role":
+        "    -r ROLE, --role=ROLE                Only show peers with role "
+        "ROLE.",
+    "shard":
+        "    -s SHARD, --shard=SHARD             Name of the Manatee shard "
+        "(cluster).",
+    "verbose": "    -v, --verbose                       Enable verbose "
+               "output.",
+    "wide": "    -w, --wide, --cinematic             Show full peernames.",
+    "zk": ("    -z ZK_IPS, --zk=ZK_IPS              The zookeeper connection "
+           "string. (e.g.,\n"
+           "                                        127.0.0.1:2181)."),
+}
+
+
+def _mk_help(summary: str, usage: str, opts: List[str]) -> str:
+    return ("%s\n\nUsage:\n    %s\n\nOptions:\n%s\n"
+            % (summary, usage, "\n".join(_HELP_OPTS[o] for o in opts)))
+
+
+COMMAND_HELP = {
+    "peers": _mk_help(
+        "Show known peers in this cluster. ",
+        "manatee-adm peers [OPTIONS]",
+        ["help", "columns", "omitHeader", "role", "shard", "zk"]),
+    "pg-status": _mk_help(
+        "Show the postgres status of this cluster. ",
+        "manatee-adm pg-status [OPTIONS] [PERIOD [COUNT]]",
+        ["help", "columns", "omitHeader", "role", "shard", "wide", "zk"]),
+    "show": _mk_help(
+        "Show cluster summary information.",
+        "manatee-adm show [OPTIONS]",
+        ["help", "shard", "verbose", "zk"]),
+    "verify": _mk_help(
+        "Verify the health of the cluster.",
+        "manatee-adm verify [OPTIONS]",
+        ["help", "verbose", "shard", "zk"]),
+}
 
 
 def _fail(msg: str) -> int:
@@ -62,7 +117,12 @@ async def _with_zk(ns, fn):
 async def _details(ns) -> det.ClusterDetails:
     fx = det.fixture_path()
     if fx:
-        return det.load_fixture(fx)
+        # shard/zk still come from flags/env for display (the reference
+        # prints opts.shard / opts.zk in "show" even with a fixture)
+        return det.load_fixture(
+            fx,
+            shard=getattr(ns, "shard", None) or os.environ.get("SHARD"),
+            zk_conn=getattr(ns, "zk", None) or os.environ.get("ZK_IPS"))
     shard = _need(ns, "shard", "SHARD", "-s/--shard")
 
     async def go(zk):
@@ -81,23 +141,40 @@ def _print_issues(cd: det.ClusterDetails, stream, leading_nl: bool) -> None:
         print("warning: %s" % w.split("\n")[0], file=stream)
 
 
-def _columns(ns, default: List[str]) -> List[str]:
+def _columns(ns, default: List[str], no_pg: bool = False) -> List[str]:
+    """ref extractColumns bin/manatee-adm:1253-1310: expand comma lists,
+    resolve aliases, reject unknown columns — and, for "peers", reject
+    pg-only columns (byte-compatible error messages)."""
+    help_text = COMMAND_HELP.get(getattr(ns, "cmd_name", ""))
     if not getattr(ns, "columns", None):
-        return default
-    cols = []
-    for chunk in ns.columns:
-        cols.extend(c.strip() for c in chunk.split(",") if c.strip())
-    for c in cols:
-        if c not in det.COLUMNS:
-            raise UsageError("unknown column %r (have: %s)"
-                             % (c, ", ".join(sorted(det.COLUMNS))))
+        cols = list(default)
+    else:
+        cols = []
+        for chunk in ns.columns:
+            for c in chunk.split(","):
+                if not c:
+                    continue
+                c = det.COLUMN_ALIASES.get(c, c)
+                if c not in det.COLUMNS:
+                    raise UsageError('unsupported column: "%s"' % c,
+                                     help_text)
+                cols.append(c)
+        if not cols:
+            raise UsageError("no columns selected", help_text)
+    if no_pg:
+        for c in cols:
+            if c in det.PG_ONLY_COLUMNS:
+                raise UsageError('column not available with this '
+                                 'subcommand: "%s"'
+                                 % det.COLUMNS[c]["label"], help_text)
     return cols
 
 
 def _check_role(ns) -> Optional[str]:
     role = getattr(ns, "role", None)
     if role and role not in det.ROLES:
-        raise UsageError('unsupported value for --role: "%s"' % role)
+        raise UsageError('unsupported value for --role: "%s"' % role,
+                         COMMAND_HELP.get(getattr(ns, "cmd_name", "")))
     return role
 
 
@@ -109,10 +186,11 @@ def cmd_version(ns) -> int:
 
 
 async def cmd_peers(ns) -> int:
+    cols = _columns(ns, det.PEERS_COLUMNS, no_pg=True)
+    role = _check_role(ns)
     cd = await _details(ns)
-    cols = _columns(ns, det.PEERS_COLUMNS)
     sys.stdout.write(det.render_table(
-        cols, cd.table_rows(cols, role=_check_role(ns)),
+        cols, cd.table_rows(cols, role=role),
         header=not ns.omit_header))
     return 0
 
@@ -121,8 +199,25 @@ async def cmd_pg_status(ns) -> int:
     cols = _columns(ns, det.STATUS_COLUMNS_WIDE if ns.wide
                     else det.STATUS_COLUMNS)
     role = _check_role(ns)
-    period = ns.period
-    count = ns.count if ns.count is not None else (None if period else 1)
+    help_text = COMMAND_HELP.get("pg-status")
+    period = count = None
+    if ns.period is not None:
+        try:
+            period = int(ns.period)
+        except ValueError:
+            period = -1
+        if period < 1:
+            raise UsageError('invalid period: "%s"' % ns.period, help_text)
+        if ns.count is not None:
+            try:
+                count = int(ns.count)
+            except ValueError:
+                count = -1
+            if count < 1:
+                raise UsageError('invalid count: "%s"' % ns.count,
+                                 help_text)
+    if count is None and period is None:
+        count = 1
     shown = 0
     while True:
         cd = await _details(ns)
@@ -591,8 +686,14 @@ def _mk_parser() -> argparse.ArgumentParser:
     sub = p.add_subparsers(dest="cmd")
 
     def add(name: str, fn, aliases=(), **kw):
+        if name in COMMAND_HELP:
+            # these subcommands print the reference's exact help text
+            kw["add_help"] = False
         sp = sub.add_parser(name, aliases=list(aliases), **kw)
-        sp.set_defaults(fn=fn)
+        sp.set_defaults(fn=fn, cmd_name=name)
+        if name in COMMAND_HELP:
+            sp.add_argument("-h", "--help", dest="want_help",
+                            action="store_true")
         sp.add_argument("-z", "--zk", help="ZooKeeper connection string "
                         "(env ZK_IPS)")
         sp.add_argument("-s", "--shard", help="shard name (env SHARD)")
@@ -613,9 +714,9 @@ def _mk_parser() -> argparse.ArgumentParser:
                     action="store_true")
     sp.add_argument("-o", "--columns", action="append")
     sp.add_argument("-r", "--role")
-    sp.add_argument("-w", "--wide", action="store_true")
-    sp.add_argument("period", nargs="?", type=int, default=None)
-    sp.add_argument("count", nargs="?", type=int, default=None)
+    sp.add_argument("-w", "--wide", "--cinematic", action="store_true")
+    sp.add_argument("period", nargs="?", default=None)
+    sp.add_argument("count", nargs="?", default=None)
 
     sp = add("show", cmd_show, help="summary + status table")
     sp.add_argument("-v", "--verbose", action="store_true")
@@ -684,6 +785,9 @@ def main(argv=None) -> int:
     if not getattr(ns, "fn", None):
         parser.print_help()
         return 2
+    if getattr(ns, "want_help", False):
+        sys.stdout.write(COMMAND_HELP[ns.cmd_name])
+        return 0
     try:
         res = ns.fn(ns)
         if asyncio.iscoroutine(res):
@@ -691,6 +795,8 @@ def main(argv=None) -> int:
         return int(res or 0)
     except UsageError as exc:
         print("manatee-adm: %s" % exc, file=sys.stderr)
+        if exc.help_text:
+            sys.stderr.write(exc.help_text)
         return 2
     except KeyboardInterrupt:
         return 130

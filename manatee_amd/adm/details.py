@@ -25,29 +25,52 @@ from ..db.waldb.client import WaldbClient
 
 TEST_STATE_ENV = "MANATEE_ADM_TEST_STATE"
 
-# column widths of the reference tables (bin/manatee-adm:1253-1420)
+# column widths of the reference tables (ref allColumns
+# bin/manatee-adm:1151-1222 — byte-compatible, incl. trailing padding)
 COLUMNS = {
     "peername": {"label": "PEERNAME", "width": 36},
     "peerabbr": {"label": "PEER", "width": 8},
     "role": {"label": "ROLE", "width": 8},
-    "ip": {"label": "IP", "width": 15},
+    "ip": {"label": "IP", "width": 16},
     "pg-online": {"label": "PG", "width": 4},
     "pg-repl": {"label": "REPL", "width": 5},
     "pg-sent": {"label": "SENT", "width": 13},
     "pg-write": {"label": "WRITE", "width": 13},
     "pg-flush": {"label": "FLUSH", "width": 13},
     "pg-replay": {"label": "REPLAY", "width": 13},
-    "pg-lag": {"label": "LAG", "width": 5},
+    "pg-lag": {"label": "LAG", "width": 6},
 }
+# ref columnAliases bin/manatee-adm:1224-1227
+COLUMN_ALIASES = {"zonename": "peername", "zoneabbr": "peerabbr"}
+PG_ONLY_COLUMNS = {"pg-online", "pg-repl", "pg-sent", "pg-write",
+                   "pg-flush", "pg-replay", "pg-lag"}
 PEERS_COLUMNS = ["role", "peername", "ip"]
 STATUS_COLUMNS = ["role", "peerabbr", "pg-online", "pg-repl", "pg-sent",
                   "pg-flush", "pg-replay", "pg-lag"]
 STATUS_COLUMNS_WIDE = ["role", "peername", "pg-online", "pg-repl", "pg-sent",
-                       "pg-write", "pg-flush", "pg-replay", "pg-lag"]
+                       "pg-flush", "pg-replay", "pg-lag"]
 ROLES = ("primary", "sync", "async", "deposed")
 
+_LAG_UNITS = {"days": 86400, "hours": 3600, "minutes": 60, "seconds": 1}
 
-def duration(seconds: Optional[float]) -> str:
+
+def lag_in_seconds(lag):
+    """ref lagInSeconds lib/adm.js:2504-2541 — a postgres interval object
+    ({days, hours, minutes, seconds}) to seconds; None for no lag;
+    the string "?" for units we cannot interpret."""
+    if not lag:
+        return None
+    if isinstance(lag, (int, float)):
+        return lag
+    seconds = 0
+    for unit, val in lag.items():
+        if unit not in _LAG_UNITS:
+            return "?"
+        seconds += val * _LAG_UNITS[unit]
+    return seconds
+
+
+def duration(seconds) -> str:
     """ref pgDuration bin/manatee-adm:1420-1437 — ``%dm%02ds``."""
     if seconds is None:
         return "-"
@@ -70,11 +93,18 @@ class PeerDetails:
         self.db_error: Optional[str] = None    # pgp_pgerr
         self.status: Optional[dict] = None     # raw waldb status
         self.repl_rows: List[dict] = []        # pg_stat_replication rows
-        self.lag_s: Optional[float] = None     # upstream replay lag
+        self.lag_s = None                      # upstream replay lag (or "?")
+
+    @property
+    def zone_id(self) -> str:
+        """The peer's name for display: zoneId when present, id otherwise
+        (ref PEERNAME = pgp_ident.zoneId, bin/manatee-adm:1379)."""
+        return self.ident.get("zoneId") or self.id
 
     @property
     def label(self) -> str:
-        return self.id[:8]
+        """ref pgp_label = zoneId.substr(0, 8) (lib/adm.js:841)."""
+        return self.zone_id[:8]
 
     @property
     def online(self) -> bool:
@@ -91,8 +121,10 @@ class PeerDetails:
         return self.repl_rows[0] if self.repl_rows else None
 
     def cell(self, col: str) -> str:
+        """ref rowForPeer bin/manatee-adm:1376-1414 (incl. the
+        ``*_lsn || *_location`` fallback for pre-PG-10 field names)."""
         if col == "peername":
-            return self.id
+            return self.zone_id
         if col == "peerabbr":
             return self.label
         if col == "role":
@@ -101,19 +133,26 @@ class PeerDetails:
             return self.ident.get("ip", "-") or "-"
         if col == "pg-online":
             return "ok" if self.online else "fail"
+        if not self.online:
+            return "-"   # ref: every pg column renders "-" on pgp_pgerr
+        if col == "pg-lag":
+            return duration(self.lag_s)
         row = self.first_repl()
+
+        def lsn(prefix: str) -> str:
+            r = row or {}
+            return r.get(prefix + "_lsn") or r.get(prefix + "_location") \
+                or "-"
         if col == "pg-repl":
             return (row or {}).get("sync_state") or "-"
         if col == "pg-sent":
-            return (row or {}).get("sent_lsn") or "-"
+            return lsn("sent")
         if col == "pg-write":
-            return (row or {}).get("write_lsn") or "-"
+            return lsn("write")
         if col == "pg-flush":
-            return (row or {}).get("flush_lsn") or "-"
+            return lsn("flush")
         if col == "pg-replay":
-            return (row or {}).get("replay_lsn") or "-"
-        if col == "pg-lag":
-            return duration(self.lag_s)
+            return lsn("replay")
         raise KeyError("unknown column %r" % col)
 
 
@@ -220,7 +259,10 @@ class ClusterDetails:
         # if the sync is down, that's all we can really check for now
         if s is None or not s.online:
             return
-        if p is not None and p.online:
+        # the primary's downstream is checked even when the primary itself
+        # is unreachable (ref loadErrors — it reports BOTH "cannot query"
+        # and "downstream not connected" for a down primary)
+        if p is not None:
             self.load_repl_errors(p, self.sync_id, "sync", self.errors)
         self.load_repl_errors(
             s, self.async_ids[0] if self.async_ids else None,
@@ -246,8 +288,18 @@ class ClusterDetails:
             out.append('peer "%s": downstream replication peer not '
                        "connected" % peer.label)
             return
-        if row.get("application_name") != ds_id:
-            ds = self.peers.get(ds_id)
+        ds = self.peers.get(ds_id)
+        if "client_addr" in row:
+            # PostgreSQL-style row: match the downstream by IP
+            # (ref loadReplErrors lib/adm.js:958-963)
+            if row.get("client_addr") != (ds.ident.get("ip") if ds
+                                          else None):
+                out.append('peer "%s": expected downstream peer to be '
+                           '"%s", but found "%s"'
+                           % (peer.label,
+                              (ds.ident.get("ip") if ds else ds_id),
+                              row.get("client_addr")))
+        elif row.get("application_name") != ds_id:
             out.append('peer "%s": expected downstream peer to be "%s", '
                        'but found "%s"'
                        % (peer.label, (ds.label if ds else ds_id),
@@ -307,15 +359,26 @@ async def load_cluster_details(zk, shard: str, *, zk_conn: str = "",
     return cd
 
 
-def load_fixture(path: str) -> ClusterDetails:
-    """Fixture seam: JSON file with
-    ``{clusterState, db: {peerId: status|null}, now?}`` — db statuses
-    replace the live queries (ref MANATEE_ADM_TEST_STATE lib/adm.js:721-745).
+def load_fixture(path: str, shard: Optional[str] = None,
+                 zk_conn: Optional[str] = None) -> ClusterDetails:
+    """Fixture seam (ref MANATEE_ADM_TEST_STATE lib/adm.js:721-745).
+
+    Accepts TWO formats:
+
+    - this repo's: ``{clusterState, db: {peerId: status|null}, now?}`` —
+      db statuses replace the live queries;
+    - the REFERENCE's internal representation (``pgs_*``/``pgp_*`` fields,
+      exactly what the reference's own golden tests feed its CLI,
+      ref test/tst.manateeAdm.js MockState) — this is what makes the
+      byte-compat test able to render reference-shaped fixtures.
     """
     with open(path) as f:
         fx = json.load(f)
-    cd = ClusterDetails(fx.get("shard", "1.fixture"), fx["clusterState"],
-                        zk_conn="UNUSED")
+    if "pgs_peers" in fx:
+        return _load_reference_fixture(fx, shard=shard, zk_conn=zk_conn)
+    cd = ClusterDetails(shard or fx.get("shard", "1.fixture"),
+                        fx["clusterState"],
+                        zk_conn=zk_conn or "UNUSED")
     dbmap = fx.get("db") or {}
     for pid, pd in cd.peers.items():
         st = dbmap.get(pid)
@@ -324,6 +387,46 @@ def load_fixture(path: str) -> ClusterDetails:
         else:
             pd.status = st
     cd._ingest_statuses(now=fx.get("now"))
+    cd.load_errors()
+    return cd
+
+
+def _load_reference_fixture(fx: dict, shard: Optional[str] = None,
+                            zk_conn: Optional[str] = None
+                            ) -> ClusterDetails:
+    """Build ClusterDetails from the reference's internal representation
+    (ref ManateeClusterDetails lib/adm.js:760-875)."""
+    peers = fx.get("pgs_peers") or {}
+
+    def ident_of(pid):
+        return (peers.get(pid) or {}).get("pgp_ident") or {"id": pid}
+
+    state = {
+        "generation": fx.get("pgs_generation"),
+        "initWal": fx.get("pgs_initwal"),
+        "oneNodeWriteMode": bool(fx.get("pgs_singleton")),
+        "primary": ident_of(fx["pgs_primary"]) if fx.get("pgs_primary")
+        else None,
+        "sync": ident_of(fx["pgs_sync"]) if fx.get("pgs_sync") else None,
+        "async": [ident_of(a) for a in fx.get("pgs_asyncs") or []],
+        "deposed": [ident_of(d) for d in fx.get("pgs_deposed") or []],
+    }
+    if fx.get("pgs_frozen"):
+        state["freeze"] = {"date": fx.get("pgs_freeze_time"),
+                           "reason": fx.get("pgs_freeze_reason")}
+    cd = ClusterDetails(shard or "1.fixture", state,
+                        zk_conn=zk_conn or "UNUSED")
+    for pid, pd in cd.peers.items():
+        raw = peers.get(pid) or {}
+        if raw.get("pgp_pgerr"):
+            pd.db_error = "connection refused"
+        else:
+            pd.status = {}
+        if raw.get("pgp_repl"):
+            pd.repl_rows = [raw["pgp_repl"]]
+        pd.lag_s = lag_in_seconds(raw.get("pgp_lag"))
+    cd.errors = list(fx.get("pgs_errors") or [])
+    cd.warnings = list(fx.get("pgs_warnings") or [])
     cd.load_errors()
     return cd
 
