@@ -9,14 +9,17 @@ server (``zkserver.py``), so the client remains usable against a real
 ZooKeeper ensemble and the on-ZK namespace/state stays protocol-compatible.
 
 All integers are big-endian.  Every frame on the wire is length-prefixed with
-an int32.  When the native extension ``manatee_amd.coord._jute`` (C++) is
-present it provides the hot encode/decode paths; this file defines the record
-layer on top and a pure-Python codec fallback selected explicitly via
-MANATEE_PURE_PY=1.
+an int32.  The primitive Writer/Reader come from the C++ extension
+``manatee_amd.native._jutec`` when it is built (``__graft_entry__.build()``
+compiles it in-tree; ``tests/test_native_jute.py`` fuzzes byte parity);
+``MANATEE_PURE_PY=1`` forces the pure-Python codec defined below, which is
+also the fallback when the extension is absent.  ``CODEC`` says which one
+is active.
 """
 
 from __future__ import annotations
 
+import os
 import struct
 from typing import List, Optional, Tuple
 
@@ -190,6 +193,23 @@ class Reader:
 
     def remaining(self) -> int:
         return len(self._buf) - self._pos
+
+
+# keep the pure-Python implementations importable under stable names
+# (the parity fuzz test compares them against the native ones)
+PyWriter = Writer
+PyReader = Reader
+
+CODEC = "python"
+if os.environ.get("MANATEE_PURE_PY") != "1":
+    try:
+        from ..native import jutec as _jutec
+    except ImportError:     # pragma: no cover - package layout issue only
+        _jutec = None
+    if _jutec is not None:
+        Writer = _jutec.Writer      # type: ignore[misc]
+        Reader = _jutec.Reader      # type: ignore[misc]
+        CODEC = "native"
 
 
 # ======================================================================

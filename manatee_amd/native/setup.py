@@ -15,6 +15,11 @@ setup(
             cxx_std=17,
             extra_compile_args=["-O3"],
         ),
+        Pybind11Extension(
+            "_jutec", ["jutec.cpp"],
+            cxx_std=17,
+            extra_compile_args=["-O3"],
+        ),
     ],
     cmdclass={"build_ext": build_ext},
 )
