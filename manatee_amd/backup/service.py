@@ -19,6 +19,8 @@ import asyncio
 import uuid as uuidmod
 from typing import Dict, List, Optional
 
+from ..common import dial
+
 from ..common.httpd import HttpServer
 from ..common.logging import Logger, null_logger
 from ..storage.provider import SnapshotStore, is_auto_snapshot
@@ -114,7 +116,7 @@ class BackupSender:
                                                            job.port),
                       size=job.size)
         reader, writer = await asyncio.wait_for(
-            asyncio.open_connection(job.host, job.port), 30)
+            dial.open_connection(job.host, job.port), 30)
         try:
             stream = await self.store.send(job.snapshot)
             async for chunk in stream:

@@ -13,6 +13,8 @@ import asyncio
 import json
 from typing import Awaitable, Callable, Dict, Optional, Tuple
 
+from . import dial
+
 from .logging import Logger, null_logger
 
 Handler = Callable[..., Awaitable[Tuple[int, object]]]
@@ -134,7 +136,7 @@ async def http_request(url: str, method: str = "GET", body: object = None,
            "content-length: %d\r\nconnection: close\r\n\r\n"
            % (method, path, hostport, len(data))).encode() + data
     reader, writer = await asyncio.wait_for(
-        asyncio.open_connection(host, int(port or 80)), timeout_s)
+        dial.open_connection(host, int(port or 80)), timeout_s)
     try:
         writer.write(req)
         await writer.drain()

@@ -23,6 +23,7 @@ import time
 from typing import Callable, Dict, List, Optional, Tuple
 
 from ..common.logging import Logger, null_logger
+from ..common import dial
 from . import jute
 from .jute import MultiOp, Reader, Stat, Writer, ZkError, ZOK, ZCONNECTIONLOSS
 
@@ -157,7 +158,7 @@ class ZkClient:
 
     async def _connect_once(self, host: str, port: int) -> None:
         reader, writer = await asyncio.wait_for(
-            asyncio.open_connection(host, port), 5.0)
+            dial.open_connection(host, port), 5.0)
         try:
             writer.write(jute.encode_connect_request(
                 self.last_zxid, self.session_timeout_ms,

@@ -16,6 +16,8 @@ import asyncio
 import struct
 from typing import Dict, List, Optional, Tuple
 
+from ..common import dial
+
 _I32 = struct.Struct(">i")
 _HDR = struct.Struct(">cI")     # type byte + length (len includes itself)
 
@@ -67,7 +69,7 @@ class PgClient:
     # ------------------------------------------------------------ lifecycle
     async def connect(self) -> None:
         self._reader, self._writer = await asyncio.wait_for(
-            asyncio.open_connection(self.host, self.port),
+            dial.open_connection(self.host, self.port),
             self.connect_timeout_s)
         params = ("user\x00%s\x00database\x00%s\x00\x00"
                   % (self.user, self.database)).encode("utf-8")

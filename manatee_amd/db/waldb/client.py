@@ -11,6 +11,8 @@ import asyncio
 import json
 from typing import Optional
 
+from ...common import dial
+
 
 class WaldbError(RuntimeError):
     pass
@@ -42,7 +44,7 @@ class WaldbClient:
     async def _ensure(self) -> None:
         if self._writer is None or self._writer.is_closing():
             self._reader, self._writer = await asyncio.wait_for(
-                asyncio.open_connection(self.host, self.port),
+                dial.open_connection(self.host, self.port),
                 self.connect_timeout_s)
 
     async def query(self, req: dict, timeout_s: Optional[float] = None

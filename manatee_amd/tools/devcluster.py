@@ -125,6 +125,12 @@ class DevPeer:
         env = dict(os.environ)
         env["PYTHONPATH"] = REPO_ROOT + os.pathsep + \
             env.get("PYTHONPATH", "")
+        if self.cluster.proxied:
+            # every outbound connection this peer (and its db child)
+            # makes is rewritten through per-directed-link proxies so
+            # tests can induce asymmetric network partitions
+            env["MANATEE_DIAL_MAP"] = os.path.join(self.dir,
+                                                   "dialmap.json")
         logf = open(os.path.join(self.dir, logname), "a")
         return subprocess.Popen(
             [sys.executable, "-m", module, "-f",
@@ -279,7 +285,8 @@ class DevCluster:
                  base_port: Optional[int] = None,
                  run_snapshotter: bool = True,
                  snapshot_interval_ms: int = 30000,
-                 snapshot_number: int = 5):
+                 snapshot_number: int = 5,
+                 proxied: bool = False):
         self.base_dir = os.path.abspath(base_dir)
         self.ip = ip
         self.engine = engine
@@ -294,6 +301,8 @@ class DevCluster:
         self.run_snapshotter = run_snapshotter
         self.snapshot_interval_ms = snapshot_interval_ms
         self.snapshot_number = snapshot_number
+        self.proxied = proxied
+        self.proxies: Dict[tuple, object] = {}
         self.zk_port = 0  # assigned below the ephemeral range in __init__
         self.zk_conn_str = ""
         self.zk_proc: Optional[subprocess.Popen] = None
@@ -375,15 +384,108 @@ class DevCluster:
                 await asyncio.sleep(0.05)
 
     async def start(self, peers: Optional[List[int]] = None) -> None:
+        if self.proxied:
+            await self._start_proxies()
         self.start_zk()
         await self.wait_zk()
         for i, peer in enumerate(self.peers):
             if peers is None or i in peers:
                 peer.start()
 
+    # ---------------------------------------------------- network partitions
+    async def _start_proxies(self) -> None:
+        """One LinkProxy per directed (src peer → dst) link: dst is
+        another peer's db/backup port or the ZK server.  Each peer's
+        dial map (MANATEE_DIAL_MAP) routes its outbound connections
+        through its own proxies, so tests can drop bytes per directed
+        link — the ipdadm-network-partition analogue
+        (ref docs/test-plan.md:24-113)."""
+        from .netproxy import LinkProxy
+        for src in self.peers:
+            pz = LinkProxy(self.ip, self.zk_port,
+                           name="peer%d->zk" % src.index)
+            await pz.start()
+            self.proxies[(src.index, "zk")] = pz
+            for dst in self.peers:
+                if dst.index == src.index:
+                    continue
+                for kind, port in (("pg", dst.pg_port),
+                                   ("backup", dst.backup_port)):
+                    p = LinkProxy(self.ip, port,
+                                  name="peer%d->peer%d:%s"
+                                  % (src.index, dst.index, kind))
+                    await p.start()
+                    self.proxies[(src.index, dst.index, kind)] = p
+            # the dial map must exist before the peer spawns
+            os.makedirs(src.dir, exist_ok=True)
+            dmap = {"%s:%d" % (self.ip, self.zk_port): pz.addr}
+            for dst in self.peers:
+                if dst.index == src.index:
+                    continue
+                dmap["%s:%d" % (self.ip, dst.pg_port)] = \
+                    self.proxies[(src.index, dst.index, "pg")].addr
+                dmap["%s:%d" % (self.ip, dst.backup_port)] = \
+                    self.proxies[(src.index, dst.index, "backup")].addr
+            with open(os.path.join(src.dir, "dialmap.json"), "w") as f:
+                json.dump(dmap, f, indent=2)
+
+    def set_link(self, a: DevPeer, b: DevPeer,
+                 drop_a2b: Optional[bool] = None,
+                 drop_b2a: Optional[bool] = None) -> None:
+        """Drop bytes traveling a→b and/or b→a, on BOTH carriers
+        (connections a initiated to b AND connections b initiated to a).
+        ``drop_a2b=True`` alone is a one-way partition: a's packets
+        never reach b, but b's still reach a."""
+        for kind in ("pg", "backup"):
+            pab = self.proxies.get((a.index, b.index, kind))
+            pba = self.proxies.get((b.index, a.index, kind))
+            if pab is not None:
+                pab.set_drops(to_server=drop_a2b, to_client=drop_b2a)
+            if pba is not None:
+                pba.set_drops(to_server=drop_b2a, to_client=drop_a2b)
+
+    def partition(self, a: DevPeer, b: DevPeer) -> None:
+        self.set_link(a, b, drop_a2b=True, drop_b2a=True)
+
+    def heal_link(self, a: DevPeer, b: DevPeer) -> None:
+        for kind in ("pg", "backup"):
+            for key in ((a.index, b.index, kind), (b.index, a.index, kind)):
+                p = self.proxies.get(key)
+                if p is not None:
+                    p.heal()
+
+    def partition_zk(self, a: DevPeer) -> None:
+        """Cut peer a off from the coordination server (its session
+        expires) while leaving every peer↔peer link intact."""
+        p = self.proxies.get((a.index, "zk"))
+        if p is not None:
+            p.set_drops(to_server=True, to_client=True)
+
+    def heal_zk(self, a: DevPeer) -> None:
+        p = self.proxies.get((a.index, "zk"))
+        if p is not None:
+            p.heal()
+
+    def isolate(self, a: DevPeer) -> None:
+        """Full network isolation of one peer (ZK + every peer link);
+        clients (the test) can still reach it directly."""
+        self.partition_zk(a)
+        for b in self.peers:
+            if b.index != a.index:
+                self.partition(a, b)
+
+    def heal_all(self) -> None:
+        for p in self.proxies.values():
+            p.heal()
+
     def stop(self) -> None:
         for peer in self.peers:
             peer.stop()
+        for p in self.proxies.values():
+            if p._server is not None:
+                p._server.close()
+            p.kill_connections()
+        self.proxies.clear()
         if self.zk_proc is not None and self.zk_proc.poll() is None:
             try:
                 os.killpg(self.zk_proc.pid, signal.SIGKILL)

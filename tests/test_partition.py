@@ -1,0 +1,205 @@
+"""Network-partition failover tests (the ipdadm tier of the reference's
+chaos plan, ref docs/test-plan.md:24-113).
+
+No netns/iptables exists in this environment, so partitions are induced
+in userspace: every outbound connection a peer makes is routed through a
+per-directed-link proxy (tools/netproxy.LinkProxy via the
+MANATEE_DIAL_MAP rewrite layer, common/dial.py), and the proxy drops
+bytes per direction.  Unlike SIGSTOP, this produces the classic
+split-brain shapes: a peer that is alive and reachable by CLIENTS but
+cut off from ZooKeeper, and asymmetric links where A hears B but B
+cannot hear A.
+
+Safety property under test, in all scenarios: once a write is
+acknowledged, it survives; and after a takeover the deposed primary can
+never acknowledge another write (the synchronous-replication gate).
+"""
+
+import asyncio
+import os
+import sys
+import time
+
+import pytest
+
+sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+
+import bench  # noqa: E402  (Writer + exact loss verification)
+from manatee_amd.db.waldb.client import WaldbClient, WaldbError  # noqa: E402
+from manatee_amd.tools.devcluster import DevCluster  # noqa: E402
+
+
+def run(coro, timeout=240):
+    return asyncio.run(asyncio.wait_for(coro, timeout))
+
+
+async def _formed(cluster_dir, shard):
+    c = DevCluster(cluster_dir, n_peers=3, shard_name=shard,
+                   session_timeout_ms=2000, proxied=True,
+                   run_snapshotter=False)
+    await c.start()
+    await c.wait_cluster(
+        lambda s: s.get("sync") and len(s.get("async", [])) == 1,
+        timeout_s=120, what="3-peer formation (proxied)")
+    await c.wait_writable(timeout_s=120)
+    w = bench.Writer(c)
+    w.start()
+    while w.seq < 30:
+        await asyncio.sleep(0.05)
+    return c, w
+
+
+async def _put_must_fail(peer, key, timeout_s=2.5):
+    """A write against `peer` must NOT be acknowledged."""
+    cli = peer.db_client()
+    try:
+        await cli.put(key, "x", timeout_s=timeout_s)
+    except (WaldbError, OSError, asyncio.TimeoutError):
+        return
+    finally:
+        await cli.close()
+    raise AssertionError("write to %s was acknowledged but must not be"
+                         % peer.id)
+
+
+def test_primary_cut_from_zk_but_not_from_clients(tmp_path):
+    """The classic split-brain shape: the primary loses ZooKeeper while
+    still running, still replicating, and still reachable by clients.
+    The cluster must promote the sync (generation+1), after which the
+    old primary must never acknowledge another write, and no
+    acknowledged write may be lost."""
+    async def go():
+        c, w = await _formed(str(tmp_path / "c"), "1.partzk")
+        try:
+            s0 = await c.cluster_state()
+            prim = c.peer_by_id(s0["primary"]["id"])
+            c.partition_zk(prim)
+
+            # session expiry (2 s) → sync takeover with a generation bump
+            s1 = await c.wait_cluster(
+                lambda s: s["generation"] > s0["generation"] and
+                s["primary"]["id"] == s0["sync"]["id"],
+                timeout_s=30, what="sync takeover after zk partition")
+            newp = await c.wait_writable(timeout_s=30)
+            assert newp.id == s0["sync"]["id"]
+
+            # the deposed primary is still alive and reachable by THIS
+            # client — but with its sync gone it must not ack anything
+            await _put_must_fail(prim, "split-brain-probe")
+
+            # every previously acknowledged write is intact
+            v = await bench.verify_no_loss(c, s1, w)
+            assert v["lost"] == 0, v
+
+            # heal: the old primary reconnects, finds itself deposed
+            c.heal_zk(prim)
+            s2 = await c.wait_cluster(
+                lambda s: any(d["id"] == prim.id
+                              for d in s.get("deposed", [])),
+                timeout_s=30, what="old primary listed as deposed")
+            assert s2["primary"]["id"] == newp.id
+
+            # rebuild it; cluster returns to full strength, still no loss
+            await c.rebuild_peer(prim)
+            await c.wait_writable(timeout_s=60)
+            s3 = await c.cluster_state()
+            v = await bench.verify_no_loss(c, s3, w)
+            assert v["lost"] == 0, v
+            await w.stop()
+        finally:
+            c.stop()
+    run(go())
+
+
+def test_sync_partitioned_from_primary_only(tmp_path):
+    """Replication link down, ZooKeeper intact on both sides: no
+    topology change may happen (liveness is ZK-based, as in the
+    reference), writes stall at the sync-commit gate rather than being
+    acknowledged unsafely, and after healing everything acked is
+    present."""
+    async def go():
+        c, w = await _formed(str(tmp_path / "c"), "1.partsync")
+        try:
+            s0 = await c.cluster_state()
+            prim = c.peer_by_id(s0["primary"]["id"])
+            sync = c.peer_by_id(s0["sync"]["id"])
+            acked_before = w.acked_count
+            c.partition(prim, sync)
+
+            # writes must STALL (no unsafe acks without the sync)
+            await asyncio.sleep(1.0)   # let in-flight acks drain
+            stall_mark = w.acked_count
+            await _put_must_fail(prim, "stall-probe")
+            await asyncio.sleep(2.0)
+            assert w.acked_count <= stall_mark + 1, \
+                "writes were acknowledged during the replication partition"
+
+            # ... and the topology must NOT change (both sessions live)
+            s1 = await c.cluster_state()
+            assert s1["generation"] == s0["generation"]
+            assert s1["primary"]["id"] == prim.id
+            assert acked_before <= w.acked_count
+
+            c.heal_link(prim, sync)
+            await c.wait_writable(timeout_s=60)
+            deadline = time.monotonic() + 30
+            while w.acked_count < stall_mark + 20:
+                assert time.monotonic() < deadline, \
+                    "writes did not resume after healing"
+                await asyncio.sleep(0.1)
+            s2 = await c.cluster_state()
+            v = await bench.verify_no_loss(c, s2, w)
+            assert v["lost"] == 0, v
+            await w.stop()
+        finally:
+            c.stop()
+    run(go())
+
+
+def test_asymmetric_partition_acks_lost_one_way(tmp_path):
+    """The shape SIGSTOP cannot produce: the sync still HEARS the
+    primary (WAL keeps arriving and applying) but the primary never
+    hears the sync's acknowledgements.  Writes must stall unacked —
+    and the probe write, though unacknowledged, is visibly present on
+    the sync, proving the asymmetry is real."""
+    async def go():
+        c, w = await _formed(str(tmp_path / "c"), "1.partasym")
+        try:
+            s0 = await c.cluster_state()
+            prim = c.peer_by_id(s0["primary"]["id"])
+            sync = c.peer_by_id(s0["sync"]["id"])
+
+            # drop only sync→primary bytes (acks); primary→sync flows
+            c.set_link(sync, prim, drop_a2b=True)
+            await asyncio.sleep(0.5)
+
+            # an un-acknowledged write...
+            await _put_must_fail(prim, "asym-probe", timeout_s=2.0)
+
+            # ...which the sync nonetheless RECEIVED (one-way link up)
+            scli = sync.db_client()
+            try:
+                got = None
+                deadline = time.monotonic() + 10
+                while got is None and time.monotonic() < deadline:
+                    got = await scli.get("asym-probe")
+                    if got is None:
+                        await asyncio.sleep(0.1)
+                assert got == "x", \
+                    "sync never received the WAL for the unacked write"
+            finally:
+                await scli.close()
+
+            # no topology change: both ZK sessions are healthy
+            s1 = await c.cluster_state()
+            assert s1["generation"] == s0["generation"]
+
+            c.heal_link(sync, prim)
+            await c.wait_writable(timeout_s=60)
+            s2 = await c.cluster_state()
+            v = await bench.verify_no_loss(c, s2, w)
+            assert v["lost"] == 0, v
+            await w.stop()
+        finally:
+            c.stop()
+    run(go())
