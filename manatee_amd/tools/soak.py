@@ -1,16 +1,20 @@
-"""Time-bounded chaos soak: a live shard under continuous synchronous
-write load with randomized faults, verifying zero acknowledged-write
-loss after every cycle.
+"""Time/cycle-bounded chaos soak: a live shard under continuous
+synchronous write load with randomized faults, verifying zero
+acknowledged-write loss after every cycle.
 
-    python -m manatee_amd.tools.soak --minutes 10 [--seed 7] [-d DIR]
+    python -m manatee_amd.tools.soak --minutes 10 [--cycles 50]
+        [--seed 7] [-d DIR]
 
 Faults drawn each cycle: SIGKILL primary / sync / async, SIGKILL just
 the database child (sitter must restart it), SIGSTOP+SIGCONT the
-primary (partition analogue).  After every fault the shard must
+primary, full-ZK outage — plus real NETWORK partitions through the
+per-link proxy layer (primary cut from ZK only, replication link cut,
+asymmetric ack loss; tools/netproxy).  After every fault the shard must
 converge back to writable with every previously-acknowledged write
-present (server-side count + recent-window readback), then the shard
-is healed to full primary/sync/async shape.  One JSON summary line on
-stdout at the end; exit 1 on any lost write or convergence failure.
+present (writer frozen, exact server-side count + recent-window
+readback), then the shard is healed to full primary/sync/async shape.
+One JSON summary line on stdout at the end; exit 1 on any lost write
+or convergence failure.
 """
 
 from __future__ import annotations
@@ -23,10 +27,12 @@ import shutil
 import sys
 import tempfile
 import time
+from typing import Optional
 
 from .devcluster import DevCluster
 
 WINDOW = 4000
+BURST = 25
 
 
 class SoakWriter:
@@ -36,11 +42,17 @@ class SoakWriter:
         self.acked_count = 0
         self.seq = 0
         self.stop_flag = False
+        self.pause_flag = False
+        self._paused = asyncio.Event()
         self.task = None
 
     async def _run(self):
         cli = None
         while not self.stop_flag:
+            if self.pause_flag:
+                self._paused.set()
+                await asyncio.sleep(0.01)
+                continue
             try:
                 if cli is None:
                     s = await self.cluster.cluster_state()
@@ -52,7 +64,7 @@ class SoakWriter:
                 # pipelined burst: high sustained write pressure
                 base = self.seq
                 items = [("soak-%d" % (base + j), base + j)
-                         for j in range(25)]
+                         for j in range(BURST)]
                 await cli.put_many(items, timeout_s=2.0)
                 for k, v in items:
                     self.window[k] = v
@@ -71,41 +83,71 @@ class SoakWriter:
     def start(self):
         self.task = asyncio.get_running_loop().create_task(self._run())
 
+    async def pause(self):
+        """Freeze at a burst boundary so the acked set is exactly
+        soak-0..seq-1 and no new acks can pad the verification count."""
+        self._paused.clear()
+        self.pause_flag = True
+        await self._paused.wait()
+
+    def resume(self):
+        self.pause_flag = False
+
     async def stop(self):
         self.stop_flag = True
+        self.pause_flag = False
         if self.task is not None:
             await self.task
 
 
 async def verify(cluster: DevCluster, writer: SoakWriter) -> int:
-    last_exc = None
-    for _attempt in range(3):
-        s = await cluster.cluster_state()
-        cli = cluster.peer_by_id(s["primary"]["id"]).db_client()
-        lost = 0
-        try:
-            present = await cli.count(prefix="soak-", timeout_s=30.0)
-            if present < writer.acked_count:
-                lost += writer.acked_count - present
-            # spot-check a sample of the recent window (the exact count
-            # above already catches any missing key; this guards values)
-            items = list(writer.window.items())
-            sample = items[-200:] + items[:50]
-            for key, val in sample:
-                if await cli.get(key) != val:
-                    lost += 1
-            return lost
-        except Exception as exc:     # mid-verify failover: retry fresh
-            last_exc = exc
-            await asyncio.sleep(1.0)
-        finally:
-            await cli.close()
-    raise RuntimeError("verification failed repeatedly: %r" % last_exc)
+    """Exact loss check: freeze the writer, count the prefix, subtract
+    any committed-but-unacked keys from the single abandoned in-flight
+    burst (they are exactly soak-seq..soak-seq+BURST-1), and read back
+    the recent window."""
+    await writer.pause()
+    try:
+        last_exc = None
+        for _attempt in range(3):
+            s = await cluster.cluster_state()
+            cli = cluster.peer_by_id(s["primary"]["id"]).db_client()
+            lost = 0
+            try:
+                present = await cli.count(prefix="soak-", timeout_s=30.0)
+                for j in range(BURST):
+                    if await cli.get("soak-%d" % (writer.seq + j)) \
+                            is not None:
+                        present -= 1
+                if present < writer.acked_count:
+                    lost += writer.acked_count - present
+                # spot-check a sample of the recent window (the exact
+                # count above already catches any missing key; this
+                # guards values)
+                items = list(writer.window.items())
+                sample = items[-200:] + items[:50]
+                for key, val in sample:
+                    if await cli.get(key) != val:
+                        lost += 1
+                return lost
+            except Exception as exc:   # mid-verify failover: retry fresh
+                last_exc = exc
+                await asyncio.sleep(1.0)
+            finally:
+                await cli.close()
+        raise RuntimeError("verification failed repeatedly: %r" % last_exc)
+    finally:
+        writer.resume()
 
 
-async def soak(minutes: float, seed: int, workdir: str) -> dict:
+ACTIONS = ["kill_primary", "kill_sync", "kill_async",
+           "kill_db_only", "pause_primary", "zk_outage",
+           "partition_zk_primary", "partition_repl", "partition_asym"]
+
+
+async def soak(minutes: float, seed: int, workdir: str,
+               cycles: Optional[int] = None) -> dict:
     rng = random.Random(seed)
-    c = DevCluster(workdir, n_peers=3, shard_name="1.soak")
+    c = DevCluster(workdir, n_peers=3, shard_name="1.soak", proxied=True)
     stats = {"cycles": 0, "kills": {}, "lost": 0, "acked": 0,
              "max_failover_s": 0.0, "failures": []}
     writer = SoakWriter(c)
@@ -120,20 +162,20 @@ async def soak(minutes: float, seed: int, workdir: str) -> dict:
             await asyncio.sleep(0.05)
 
         deadline = time.monotonic() + minutes * 60.0
-        while time.monotonic() < deadline:
+        while time.monotonic() < deadline and \
+                (cycles is None or stats["cycles"] < cycles):
             s = await c.cluster_state()
-            action = rng.choice(
-                ["kill_primary", "kill_sync", "kill_async",
-                 "kill_db_only", "pause_primary", "zk_outage"])
+            action = rng.choice(ACTIONS)
             stats["kills"][action] = stats["kills"].get(action, 0) + 1
             prim = c.peer_by_id(s["primary"]["id"])
+            sync = c.peer_by_id(s["sync"]["id"])
             victim = None
             t0 = time.monotonic()
             if action == "kill_primary":
                 victim = prim
                 victim.kill9()
             elif action == "kill_sync":
-                victim = c.peer_by_id(s["sync"]["id"])
+                victim = sync
                 victim.kill9()
             elif action == "kill_async":
                 victim = c.peer_by_id(s["async"][0]["id"])
@@ -147,6 +189,27 @@ async def soak(minutes: float, seed: int, workdir: str) -> dict:
                 await asyncio.sleep(rng.uniform(1.0, 4.0))
                 c.start_zk()
                 await c.wait_zk()
+            elif action == "partition_zk_primary":
+                # primary cut from ZK only: session expiry → sync
+                # takeover; the old primary is deposed on heal
+                c.partition_zk(prim)
+                try:
+                    await c.wait_cluster(
+                        lambda st: st["generation"] > s["generation"],
+                        timeout_s=30, what="takeover under zk partition")
+                finally:
+                    c.heal_zk(prim)
+            elif action == "partition_repl":
+                # replication link down, ZK intact: writes stall but
+                # topology must hold; then heal
+                c.partition(prim, sync)
+                await asyncio.sleep(rng.uniform(2.0, 4.0))
+                c.heal_link(prim, sync)
+            elif action == "partition_asym":
+                # one-way: the primary stops hearing the sync's acks
+                c.set_link(sync, prim, drop_a2b=True)
+                await asyncio.sleep(rng.uniform(2.0, 4.0))
+                c.heal_link(sync, prim)
 
             try:
                 await c.wait_writable(timeout_s=60)
@@ -211,13 +274,17 @@ async def soak(minutes: float, seed: int, workdir: str) -> dict:
 
 def main(argv=None) -> int:
     ap = argparse.ArgumentParser(prog="manatee-soak")
-    ap.add_argument("--minutes", type=float, default=5.0)
+    ap.add_argument("--minutes", type=float, default=5.0,
+                    help="hard wall-clock budget")
+    ap.add_argument("--cycles", type=int, default=None,
+                    help="stop after N fault cycles (within the budget)")
     ap.add_argument("--seed", type=int, default=0)
     ap.add_argument("-d", "--dir", default=None)
     ns = ap.parse_args(argv)
     workdir = ns.dir or tempfile.mkdtemp(prefix="manatee-soak-")
     try:
-        stats = asyncio.run(soak(ns.minutes, ns.seed, workdir))
+        stats = asyncio.run(soak(ns.minutes, ns.seed, workdir,
+                                 cycles=ns.cycles))
     finally:
         if ns.dir is None:
             shutil.rmtree(workdir, ignore_errors=True)

@@ -28,6 +28,7 @@ def run(coro, timeout=600):
     return asyncio.run(asyncio.wait_for(coro, timeout))
 
 
+@pytest.mark.timeout(1000)
 def test_chaos_random_kills_zero_acked_loss(tmp_path):
     """Randomized SIGKILL chaos under continuous synchronously-replicated
     write load (the docs/test-plan.md tier): every cycle kills a random
@@ -102,6 +103,32 @@ def test_chaos_random_kills_zero_acked_loss(tmp_path):
     run(go(), timeout=900)
 
 
+@pytest.mark.timeout(1800)
+def test_soak_50_cycles_zero_acked_loss():
+    """Driver-attested chaos depth: a ≥50-cycle randomized soak
+    (SIGKILLs of every role, db-child kill, SIGSTOP, full-ZK outage,
+    and real network partitions through the per-link proxy layer —
+    including the asymmetric shapes) under pipelined synchronous write
+    load, zero acknowledged-write loss, exact verification after every
+    cycle (tools/soak.py)."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, "-m", "manatee_amd.tools.soak",
+         "--cycles", "50", "--minutes", "20", "--seed", "42"],
+        capture_output=True, text=True, timeout=1500, env=env, cwd=REPO)
+    tail = "\n".join(r.stderr.splitlines()[-12:])
+    assert r.returncode == 0, tail + "\n" + r.stdout
+    stats = json.loads(r.stdout.splitlines()[-1])
+    assert stats["ok"], stats
+    assert stats["cycles"] >= 50, stats
+    assert stats["lost"] == 0, stats
+    assert not stats["failures"], stats
+    # every fault family must actually have fired across 50 cycles
+    assert len(stats["kills"]) >= 7, stats["kills"]
+
+
+@pytest.mark.timeout(700)
 def test_bench_contract_runs_and_reports():
     """bench.py must emit one valid JSON line with the BASELINE.json
     metric and zero acknowledged-write loss."""
@@ -136,3 +163,10 @@ def test_native_codec_is_loaded_and_used():
     payload = b"x" * 1000
     frame = codec.encode_frame(payload)
     assert list(walmod.parse_frames(frame)) == [(len(frame), payload)]
+    # same for the native jute wire codec: the ZK client/server must be
+    # running over the C++ primitives, not the pure-Python fallback
+    from manatee_amd.coord import jute
+    from manatee_amd.native import jutec
+    assert jutec is not None, "native _jutec extension missing on the box"
+    assert jute.CODEC == "native"
+    assert jute.Writer is jutec.Writer
