@@ -241,11 +241,13 @@ async def one_failover(cluster: DevCluster, writer: Writer) -> dict:
             "lost_acked_writes": v["lost"], "acked_checked": v["checked"]}
 
 
-async def run_rank(rank: int, steps: int, warmup: int, base_dir: str
-                   ) -> dict:
+async def run_rank(rank: int, steps: int, warmup: int, base_dir: str,
+                   engine: str = "waldb",
+                   session_timeout_ms: int = SESSION_TIMEOUT_MS) -> dict:
     cluster = DevCluster(os.path.join(base_dir, "rank%d" % rank),
                          n_peers=3, shard_name="%d.bench" % (rank + 1),
-                         session_timeout_ms=SESSION_TIMEOUT_MS)
+                         engine=engine,
+                         session_timeout_ms=session_timeout_ms)
     results = []
     total_lost = 0
     total_checked = 0
@@ -300,6 +302,15 @@ def main() -> int:
     ap.add_argument("--steps", type=int, default=5)
     ap.add_argument("--warmup", type=int, default=1)
     ap.add_argument("--workdir", default=None)
+    ap.add_argument("--engine", choices=("waldb", "postgres"),
+                    default="waldb",
+                    help="postgres = the engine=postgres management path "
+                         "(minipg binaries, libpq writes)")
+    ap.add_argument("--session-timeout-ms", type=int,
+                    default=SESSION_TIMEOUT_MS,
+                    help="ZK session timeout (failure-detection bound); "
+                         "2000 matches the reference's test tier, 60000 "
+                         "its production tier")
     ns = ap.parse_args()
 
     rank = int(os.environ.get("RANK", "0"))
@@ -307,7 +318,9 @@ def main() -> int:
 
     base_dir = ns.workdir or tempfile.mkdtemp(prefix="manatee-bench-")
     try:
-        res = asyncio.run(run_rank(rank, ns.steps, ns.warmup, base_dir))
+        res = asyncio.run(run_rank(
+            rank, ns.steps, ns.warmup, base_dir, engine=ns.engine,
+            session_timeout_ms=ns.session_timeout_ms))
     finally:
         if ns.workdir is None:   # keep logs when an explicit workdir is given
             shutil.rmtree(base_dir, ignore_errors=True)
@@ -349,12 +362,12 @@ def main() -> int:
         "data": "synthetic",
         "config": {
             "model": "manatee shard (primary/sync/async)",
-            "engine": "waldb",
+            "engine": ns.engine,
             "peers_per_shard": 3,
             "shards": world,
             "parallelism": "one shard per rank",
             "workload": "continuous synchronously-replicated writes",
-            "session_timeout_ms": SESSION_TIMEOUT_MS,
+            "session_timeout_ms": ns.session_timeout_ms,
             "kill_mode": "SIGKILL of primary sitter+db+backupserver",
             "p99_s": round(res["p99"], 4),
             "mean_s": round(res["mean"], 4),

@@ -66,7 +66,7 @@ class DevPeer:
     # ------------------------------------------------------------- configs
     def sitter_config(self) -> dict:
         c = self.cluster
-        return {
+        out = {
             "ip": self.ip,
             "postgresPort": self.pg_port,
             "backupPort": self.backup_port,
@@ -88,6 +88,16 @@ class DevPeer:
                 "oneNodeWriteMode": c.singleton,
             },
         }
+        if c.engine == "postgres":
+            # versioned binary dirs under pgBaseDir (minipg
+            # initdb/postgres shims, ref resolveVersionedPaths
+            # lib/postgresMgr.js:569-634)
+            cfg = out["postgresMgrCfg"]
+            cfg["versions"] = {"12": "12.0"}
+            cfg["defaultVersion"] = "12"
+            cfg["pgBaseDir"] = c.pg_base_dir
+            cfg["dbUser"] = "postgres"
+        return out
 
     def backupserver_config(self) -> dict:
         return {
@@ -161,6 +171,7 @@ class DevPeer:
         pids = []
         candidates = [os.path.join(data, "db_child.pid"),
                       os.path.join(data, "waldb.pid"),
+                      os.path.join(data, "postmaster.pid"),
                       # survives dataset replacement by restores
                       os.path.join(self.store_dir, "db_child.pid")]
         for path in candidates:
@@ -175,8 +186,8 @@ class DevPeer:
                         "utf-8", "replace")
             except OSError:
                 continue        # no such process
-            if "waldb" in cmdline and data in cmdline \
-                    and pid not in pids:
+            if ("waldb" in cmdline or "postgres" in cmdline) \
+                    and data in cmdline and pid not in pids:
                 pids.append(pid)
         return pids
 
@@ -260,7 +271,10 @@ class DevPeer:
             self.sitter_proc.poll() is None
 
     # -------------------------------------------------------------- clients
-    def db_client(self) -> WaldbClient:
+    def db_client(self):
+        if self.cluster.engine == "postgres":
+            from ..db.pgkv import PgKvClient
+            return PgKvClient(self.ip, self.pg_port)
         return WaldbClient(self.ip, self.pg_port)
 
     async def http_status(self, path: str = "/state"):
@@ -303,6 +317,9 @@ class DevCluster:
         self.snapshot_number = snapshot_number
         self.proxied = proxied
         self.proxies: Dict[tuple, object] = {}
+        self.pg_base_dir = ""
+        if engine == "postgres":
+            self.pg_base_dir = self._write_minipg_binaries()
         self.zk_port = 0  # assigned below the ephemeral range in __init__
         self.zk_conn_str = ""
         self.zk_proc: Optional[subprocess.Popen] = None
@@ -346,6 +363,28 @@ class DevCluster:
         peer = DevPeer(self, len(self.peers), base)
         self.peers.append(peer)
         return peer
+
+    def _write_minipg_binaries(self) -> str:
+        """Install minipg's initdb/postgres shims under the versioned
+        layout the engine expects: <pgBaseDir>/<version>/bin/{initdb,
+        postgres} (ref resolveVersionedPaths lib/postgresMgr.js:569-634;
+        the reference's mkdevsitters builds real PG from source the same
+        shape)."""
+        base = os.path.join(self.base_dir, "pgbase")
+        for version in ("12.0",):
+            bindir = os.path.join(base, version, "bin")
+            os.makedirs(bindir, exist_ok=True)
+            for name, fn in (("initdb", "initdb_main"),
+                             ("postgres", "postgres_main")):
+                path = os.path.join(bindir, name)
+                with open(path, "w") as f:
+                    f.write(
+                        "#!%s\nimport sys\nsys.path.insert(0, %r)\n"
+                        "from manatee_amd.db.minipg.server import %s\n"
+                        "sys.exit(%s(%r, sys.argv[1:]))\n"
+                        % (sys.executable, REPO_ROOT, fn, fn, version))
+                os.chmod(path, 0o755)
+        return base
 
     # -------------------------------------------------------------- control
     def start_zk(self) -> None:

@@ -152,6 +152,26 @@ def test_bench_contract_runs_and_reports():
     assert out["ms_per_step"] * out["steps"] / 1000.0 <= elapsed + 1
 
 
+@pytest.mark.timeout(700)
+def test_bench_postgres_engine_failover():
+    """The engine=postgres management path (minipg binaries, libpq
+    writes, pg_stat_replication gating) through the same failover
+    benchmark: zero acknowledged-write loss, well inside the reference's
+    30 s convergence bound."""
+    env = dict(os.environ)
+    env["PYTHONPATH"] = REPO + os.pathsep + env.get("PYTHONPATH", "")
+    r = subprocess.run(
+        [sys.executable, os.path.join(REPO, "bench.py"),
+         "--engine", "postgres", "--steps", "3", "--warmup", "1"],
+        capture_output=True, text=True, timeout=600, env=env, cwd=REPO)
+    assert r.returncode == 0, r.stdout + r.stderr
+    line = [l for l in r.stdout.splitlines() if l.startswith("{")][-1]
+    out = json.loads(line)
+    assert out["config"]["engine"] == "postgres"
+    assert out["config"]["acked_writes_lost"] == 0
+    assert out["value"] < 30.0, "failover slower than the reference bound"
+
+
 def test_native_codec_is_loaded_and_used():
     """On the dedicated box the in-tree C++ codec must be present (the
     snapshot carries the built .so) and wal.py must be using it — the
