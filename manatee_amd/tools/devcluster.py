@@ -64,6 +64,24 @@ class DevPeer:
         self.snap_proc: Optional[subprocess.Popen] = None
 
     # ------------------------------------------------------------- configs
+    def storage_cfg(self) -> dict:
+        c = self.cluster
+        if c.storage_provider == "zfs":
+            # one fakezfs pool per peer (each peer = its own host in the
+            # reference's deployment model); the shim bakes in the state
+            # root since zfs invocations run env-scrubbed
+            from .fakezfs import install_fakezfs
+            zfs_path = install_fakezfs(os.path.join(self.dir, "zfsbin"),
+                                       os.path.join(self.dir, "zfspool"))
+            for parent in ("tank", "tank/manatee"):
+                subprocess.run([zfs_path, "create", "-o", "canmount=off",
+                                parent], capture_output=True)
+            return {"provider": "zfs",
+                    "dataset": "tank/manatee/data",
+                    "mountpoint": os.path.join(self.store_dir, "live"),
+                    "zfsPath": zfs_path}
+        return {"provider": "dir", "mountpoint": self.store_dir}
+
     def sitter_config(self) -> dict:
         c = self.cluster
         out = {
@@ -78,8 +96,7 @@ class DevPeer:
             },
             "postgresMgrCfg": {
                 "engine": c.engine,
-                "storageCfg": {"provider": "dir",
-                               "mountpoint": os.path.join(self.store_dir)},
+                "storageCfg": self.storage_cfg(),
                 "healthChkInterval": c.health_interval_ms,
                 "healthChkTimeout": c.health_timeout_ms,
                 "opsTimeout": c.ops_timeout_ms,
@@ -104,16 +121,14 @@ class DevPeer:
             "ip": self.ip,
             "backupServerCfg": {"port": self.backup_port},
             "backupSenderCfg": {
-                "storageCfg": {"provider": "dir",
-                               "mountpoint": os.path.join(self.store_dir)},
+                "storageCfg": self.storage_cfg(),
             },
         }
 
     def snapshotter_config(self) -> dict:
         c = self.cluster
         return {
-            "storageCfg": {"provider": "dir",
-                           "mountpoint": os.path.join(self.store_dir)},
+            "storageCfg": self.storage_cfg(),
             "pollInterval": c.snapshot_interval_ms,
             "snapshotNumber": c.snapshot_number,
             "statusUrl": "http://%s:%d/ping" % (self.ip,
@@ -300,7 +315,8 @@ class DevCluster:
                  run_snapshotter: bool = True,
                  snapshot_interval_ms: int = 30000,
                  snapshot_number: int = 5,
-                 proxied: bool = False):
+                 proxied: bool = False,
+                 storage_provider: str = "dir"):
         self.base_dir = os.path.abspath(base_dir)
         self.ip = ip
         self.engine = engine
@@ -317,6 +333,7 @@ class DevCluster:
         self.snapshot_number = snapshot_number
         self.proxied = proxied
         self.proxies: Dict[tuple, object] = {}
+        self.storage_provider = storage_provider
         self.pg_base_dir = ""
         if engine == "postgres":
             self.pg_base_dir = self._write_minipg_binaries()
@@ -574,7 +591,14 @@ class DevCluster:
         async."""
         import shutil
         from ..adm import core as adm
+        from ..storage import open_store
         peer.kill9()
+        # destroy through the provider (for zfs the pool state lives
+        # outside store_dir; ref deposed ⇒ destroyDataset lib/adm.js:1479)
+        try:
+            await open_store(peer.storage_cfg(), log=None).destroy()
+        except Exception:
+            pass
         shutil.rmtree(peer.store_dir, ignore_errors=True)
         zk = await adm.create_zk_client(self.zk_conn_str)
         try:
