@@ -240,6 +240,13 @@ class DbManager:
         self.online = False
         self.healthy = False
         self.writable = False
+        # drop any cached engine connection: it points at the process
+        # being killed, and a stale-but-"connected" client would feed
+        # EOFs to the first probes against the next incarnation
+        try:
+            await self.engine.close()
+        except Exception:
+            pass
         if proc.returncode is None:
             try:
                 await procutil.kill_escalate(proc.pid, self.ops_timeout_s,

@@ -338,6 +338,25 @@ class MinipgServer(WaldbServer):
             return [self._row_desc(["t"]), self._data_row([val]),
                     self._complete("SELECT 1")]
 
+        if "pg_stat_wal_receiver" in low:
+            # the standby-side receiver status (real PG ≥9.6): one row
+            # while this server is a standby.  'diverged' is a minipg
+            # extension (real PG surfaces divergence only in its log);
+            # the manager uses it to trigger the restore-on-divergence
+            # path (ref standby-failure ⇒ restore :1339-1373).
+            cols = ["status", "conninfo"]
+            out = [self._row_desc(cols)]
+            if self.role == "standby":
+                status = {"streaming": "streaming",
+                          "diverged": "diverged"}.get(
+                              self.upstream_status, "stopped")
+                out.append(self._data_row(
+                    [status, "host=%s" % (self.upstream or "")]))
+                out.append(self._complete("SELECT 1"))
+            else:
+                out.append(self._complete("SELECT 0"))
+            return out
+
         if "pg_stat_replication" in low:
             return self._stat_replication()
 

@@ -113,9 +113,24 @@ class PgClient:
     # --------------------------------------------------------------- query
     async def query(self, sql: str,
                     timeout_s: float = 30.0) -> PgResult:
-        """Simple-protocol query; returns the LAST result set."""
+        """Simple-protocol query; returns the LAST result set.  Any
+        transport-level failure (EOF from a dead backend, timeout,
+        reset) tears the connection down so the next use reconnects —
+        a stale cached connection must never masquerade as a live one."""
         async with self._lock:   # serialized, ref :1990-2172
-            return await asyncio.wait_for(self._query(sql), timeout_s)
+            try:
+                return await asyncio.wait_for(self._query(sql), timeout_s)
+            except (asyncio.IncompleteReadError, EOFError,
+                    ConnectionError, OSError, asyncio.TimeoutError):
+                writer = self._writer
+                self._writer = None
+                self._reader = None
+                if writer is not None:
+                    try:
+                        writer.close()
+                    except Exception:
+                        pass
+                raise
 
     async def _query(self, sql: str) -> PgResult:
         if self._writer is None:

@@ -354,7 +354,11 @@ class PostgresEngine(Engine):
             cli = await asyncio.wait_for(self._connected_cli(), timeout_s)
             await cli.query("SELECT current_time;", timeout_s=timeout_s)
             return True
-        except (PgError, OSError, asyncio.TimeoutError, ConnectionError):
+        except (PgError, OSError, asyncio.TimeoutError, ConnectionError,
+                asyncio.IncompleteReadError, EOFError):
+            # IncompleteReadError is an EOFError, NOT a ConnectionError:
+            # missing it here once turned a routine post-restart ping
+            # into a spurious full restore on every failover
             await self.close()
             return False
 
@@ -392,18 +396,35 @@ class PostgresEngine(Engine):
         cur = await cli.query(q["last_replay_lsn" if in_recovery
                                 else "current_lsn"])
         lrt = None
+        upstream_status = "n/a"
         if in_recovery:
             r = await cli.query(
                 "SELECT extract(epoch from "
                 "pg_last_xact_replay_timestamp()) as t;")
             if r.rows and r.rows[0][0] is not None:
                 lrt = float(r.rows[0][0])
+            # receiver status (PG ≥9.6 pg_stat_wal_receiver): the
+            # manager's streaming/diverged verdicts depend on it
+            # (ref _await_streaming / restore-on-failure :1339-1373)
+            try:
+                rr = await cli.query(
+                    "SELECT status FROM pg_stat_wal_receiver;")
+                if rr.rows:
+                    st = rr.rows[0][0]
+                    upstream_status = ("streaming" if st == "streaming"
+                                       else "diverged" if st == "diverged"
+                                       else "disconnected")
+                else:
+                    upstream_status = "disconnected"
+            except PgError:
+                upstream_status = "disconnected"
         return {
             "ok": True,
             "role": "standby" if in_recovery else "primary",
             "current_lsn": cur.rows[0][0],
             "replay_lsn": cur.rows[0][0] if in_recovery else None,
             "last_replay_time": lrt,
+            "upstream_status": upstream_status,
             "replication": repl,
         }
 
