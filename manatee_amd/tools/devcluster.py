@@ -110,8 +110,8 @@ class DevPeer:
             # initdb/postgres shims, ref resolveVersionedPaths
             # lib/postgresMgr.js:569-634)
             cfg = out["postgresMgrCfg"]
-            cfg["versions"] = {"12": "12.0"}
-            cfg["defaultVersion"] = "12"
+            cfg["versions"] = {"12": "12.0", "9.6": "9.6.3"}
+            cfg["defaultVersion"] = c.pg_version
             cfg["pgBaseDir"] = c.pg_base_dir
             cfg["dbUser"] = "postgres"
         return out
@@ -316,7 +316,8 @@ class DevCluster:
                  snapshot_interval_ms: int = 30000,
                  snapshot_number: int = 5,
                  proxied: bool = False,
-                 storage_provider: str = "dir"):
+                 storage_provider: str = "dir",
+                 pg_version: str = "12"):
         self.base_dir = os.path.abspath(base_dir)
         self.ip = ip
         self.engine = engine
@@ -334,6 +335,7 @@ class DevCluster:
         self.proxied = proxied
         self.proxies: Dict[tuple, object] = {}
         self.storage_provider = storage_provider
+        self.pg_version = pg_version
         self.pg_base_dir = ""
         if engine == "postgres":
             self.pg_base_dir = self._write_minipg_binaries()
@@ -388,7 +390,7 @@ class DevCluster:
         the reference's mkdevsitters builds real PG from source the same
         shape)."""
         base = os.path.join(self.base_dir, "pgbase")
-        for version in ("12.0",):
+        for version in ("12.0", "9.6.3"):
             bindir = os.path.join(base, version, "bin")
             os.makedirs(bindir, exist_ok=True)
             for name, fn in (("initdb", "initdb_main"),

@@ -214,3 +214,132 @@ def test_adm_against_live_pg_cluster(tmp_path):
         finally:
             c.stop()
     run(go())
+
+
+def test_pg96_recovery_conf_mode(tmp_path):
+    """The pre-12 path: recovery.conf with standby_mode=on +
+    trigger_file, xlog/location query spellings, *_location columns in
+    pg_stat_replication (ref resolveWalTranslations :649-677,
+    _updateUpstreamConf :2188-2274)."""
+    async def go():
+        c = DevCluster(str(tmp_path / "c"), n_peers=3,
+                       shard_name="1.pg96", engine="postgres",
+                       pg_version="9.6", run_snapshotter=False)
+        try:
+            await c.start()
+            s = await c.wait_cluster(
+                lambda s: s.get("sync") and len(s.get("async", [])) == 1,
+                timeout_s=120, what="formation on 9.6")
+            prim = await c.wait_writable(timeout_s=120)
+
+            # the sync runs with a recovery.conf, not standby.signal
+            syncp = c.peer_by_id(s["sync"]["id"])
+            data = os.path.join(syncp.store_dir, "live", "data")
+            assert os.path.exists(os.path.join(data, "recovery.conf"))
+            assert not os.path.exists(os.path.join(data, "standby.signal"))
+            with open(os.path.join(data, "PG_VERSION")) as f:
+                assert f.read().strip() == "9.6"
+
+            # 9.x query spellings over libpq
+            cli = PgClient(prim.ip, prim.pg_port, "postgres")
+            await cli.connect()
+            r = await cli.query(
+                "SELECT pg_current_xlog_location() as loc;")
+            assert "/" in r.rows[0][0]
+            r = await cli.query("SELECT * FROM pg_stat_replication;")
+            assert "sent_location" in r.columns
+            assert "sent_lsn" not in r.columns
+            await cli.close()
+
+            # failover works through the recovery.conf trigger_file path
+            kcli = prim.db_client()
+            for i in range(20):
+                await kcli.put("v%d" % i, i)
+            await kcli.close()
+            prim.kill9()
+            await c.wait_cluster(
+                lambda st: st["generation"] > s["generation"],
+                timeout_s=60, what="9.6 takeover")
+            newp = await c.wait_writable(timeout_s=60)
+            ncli = newp.db_client()
+            assert await ncli.count(prefix="v") == 20
+            await ncli.close()
+        finally:
+            c.stop()
+    run(go())
+
+
+def test_diverged_standby_reports_via_wal_receiver(tmp_path):
+    """A standby whose history does not match its upstream (separate
+    initdb → different system identity) must report
+    pg_stat_wal_receiver.status = 'diverged' — the signal the manager's
+    restore-on-divergence path consumes (ref standby failure ⇒ full
+    restore, lib/postgresMgr.js:1339-1373)."""
+    import subprocess
+    import sys as _sys
+    from manatee_amd.common import confparser
+
+    async def go():
+        base = tmp_path
+        bindir = None
+        c = DevCluster(str(base / "scratch"), n_peers=0,
+                       engine="postgres", run_snapshotter=False)
+        bindir = os.path.join(c.pg_base_dir, "12.0", "bin")
+
+        def initdb(d):
+            r = subprocess.run([os.path.join(bindir, "initdb"), "-D", d],
+                               capture_output=True, text=True)
+            assert r.returncode == 0, r.stderr
+
+        prim_dir = str(base / "prim")
+        stby_dir = str(base / "stby")
+        initdb(prim_dir)
+        initdb(stby_dir)       # separate identity ⇒ diverged from prim
+        pport, sport = c.peers or None, None  # unused; pick free ports
+        import socket
+
+        def free():
+            s = socket.socket()
+            s.bind(("127.0.0.1", 0))
+            p = s.getsockname()[1]
+            s.close()
+            return p
+        pport, sport = free(), free()
+        confparser.write(os.path.join(prim_dir, "postgresql.conf"),
+                         {"listen_addresses": "'127.0.0.1'",
+                          "port": str(pport)})
+        confparser.write(os.path.join(stby_dir, "postgresql.conf"),
+                         {"listen_addresses": "'127.0.0.1'",
+                          "port": str(sport),
+                          "primary_conninfo":
+                          "'host=127.0.0.1 port=%d user=postgres "
+                          "application_name=stby'" % pport})
+        open(os.path.join(stby_dir, "standby.signal"), "w").close()
+        procs = []
+        try:
+            for d in (prim_dir, stby_dir):
+                procs.append(subprocess.Popen(
+                    [os.path.join(bindir, "postgres"), "-D", d],
+                    stdout=subprocess.DEVNULL,
+                    stderr=subprocess.DEVNULL))
+            scli = PgClient("127.0.0.1", sport, "postgres")
+            deadline = time.monotonic() + 20
+            status = None
+            while time.monotonic() < deadline:
+                try:
+                    if not scli.connected:
+                        await scli.connect()
+                    r = await scli.query(
+                        "SELECT status FROM pg_stat_wal_receiver;")
+                    if r.rows and r.rows[0][0] == "diverged":
+                        status = "diverged"
+                        break
+                except Exception:
+                    await scli.close()
+                await asyncio.sleep(0.2)
+            assert status == "diverged", "receiver never reported diverged"
+            await scli.close()
+        finally:
+            for p in procs:
+                p.kill()
+    run(go())

@@ -126,3 +126,34 @@ def test_zfs_provider_bootstrap_under_write_load(tmp_path):
         finally:
             c.stop()
     run(go())
+
+
+def test_zfs_provider_with_postgres_engine(tmp_path):
+    """The reference's actual deployment shape: the postgres engine on
+    ZFS datasets (conf per major, zfs snapshots, stream bootstrap) —
+    formation + failover + zero loss with both substrates in play."""
+    async def go():
+        c = DevCluster(str(tmp_path / "c"), n_peers=3,
+                       shard_name="1.zfspg", storage_provider="zfs",
+                       engine="postgres", run_snapshotter=False)
+        try:
+            await c.start()
+            s = await c.wait_cluster(
+                lambda s: s.get("sync") and len(s.get("async", [])) == 1,
+                timeout_s=120, what="formation (zfs + postgres)")
+            prim = await c.wait_writable(timeout_s=120)
+            cli = prim.db_client()
+            for i in range(25):
+                await cli.put("zp%d" % i, i)
+            await cli.close()
+            prim.kill9()
+            await c.wait_cluster(
+                lambda st: st["generation"] > s["generation"],
+                timeout_s=60, what="takeover (zfs + postgres)")
+            newp = await c.wait_writable(timeout_s=60)
+            cli = newp.db_client()
+            assert await cli.count(prefix="zp") == 25
+            await cli.close()
+        finally:
+            c.stop()
+    run(go())
