@@ -304,30 +304,42 @@ async def cmd_status(ns) -> int:
 
 
 def _status_obj(cd: det.ClusterDetails) -> dict:
-    def peer_obj(pid: str) -> dict:
+    """The reference's `status` shape (ref _formatState lib/adm.js:475-502
+    + _addPostgresStatus :348-427): per-shard keys __FROZEN__ ("date:
+    reason"), primary/sync, asyncN and deposedN (suffix empty for the
+    first), each peer = its cluster-state identity + online/repl/lag/
+    error from the live probes (deposed peers are not probed)."""
+    def peer_obj(pid: str, probed: bool = True) -> dict:
         pd = cd.peers[pid]
         o = dict(pd.ident)
+        if not probed:
+            return o
         o["online"] = pd.online
         if pd.db_error:
             o["error"] = pd.db_error
-        row = pd.first_repl()
-        if row:
-            o["repl"] = row
-        if pd.lag_s is not None:
-            o["lag_s"] = round(pd.lag_s, 3)
+        if pd.online:
+            o["repl"] = pd.first_repl() or {}
+            if pd.role != "primary":
+                lag_s = pd.lag_s if isinstance(pd.lag_s, (int, float)) \
+                    else None
+                o["lag"] = {"time_lag": {
+                    "minutes": int(lag_s // 60),
+                    "seconds": int(lag_s % 60),
+                }} if lag_s is not None else {}
         return o
-    out = {
-        "generation": cd.generation,
-        "initWal": cd.init_wal,
-        "oneNodeWriteMode": cd.singleton,
-        "freeze": cd.freeze,
-        "primary": peer_obj(cd.primary_id) if cd.primary_id else None,
-        "sync": peer_obj(cd.sync_id) if cd.sync_id else None,
-        "async": [peer_obj(a) for a in cd.async_ids],
-        "deposed": [peer_obj(d) for d in cd.deposed_ids],
-        "errors": cd.errors,
-        "warnings": cd.warnings,
-    }
+    out = {}
+    if cd.freeze:
+        out["__FROZEN__"] = "%s: %s" % (cd.freeze.get("date"),
+                                        cd.freeze.get("reason"))
+    if cd.primary_id:
+        out["primary"] = peer_obj(cd.primary_id)
+    if cd.sync_id:
+        out["sync"] = peer_obj(cd.sync_id)
+    for i, aid in enumerate(cd.async_ids):
+        out["async" + ("" if i == 0 else str(i))] = peer_obj(aid)
+    for i, did in enumerate(cd.deposed_ids):
+        out["deposed" + ("" if i == 0 else str(i))] = \
+            peer_obj(did, probed=False)
     return out
 
 

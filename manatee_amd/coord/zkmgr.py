@@ -117,6 +117,7 @@ class ZkMgr:
         self._listeners: Dict[str, List[Callable]] = {}
         self._tasks: "asyncio.Queue[Optional[Callable]]" = asyncio.Queue()
         self._worker: Optional[asyncio.Task] = None
+        self._resync_task: Optional[asyncio.Task] = None
         self._my_election_node: Optional[str] = None
 
     # --------------------------------------------------------------- events
@@ -134,9 +135,26 @@ class ZkMgr:
     async def init(self) -> None:
         """Connect, set up the namespace, join the election, arm watches,
         then emit ``init`` (ref init/setupData :412-586)."""
-        self._worker = asyncio.get_running_loop().create_task(
-            self._work_loop())
+        loop = asyncio.get_running_loop()
+        self._worker = loop.create_task(self._work_loop())
+        self._resync_task = loop.create_task(self._resync_loop())
         await self._setup_client()
+
+    async def _resync_loop(self) -> None:
+        """Low-frequency watch-loss safety net: re-list the election and
+        re-read the state every few seconds.  Both handlers debounce
+        (no event is emitted unless something actually changed) and
+        re-arm their one-shot watches, so ANY lost watch — a server
+        bug, a dropped notification, a race this code has not imagined
+        — heals within one period instead of blinding the peer until
+        session expiry.  Liveness here is the product; purity of the
+        watch discipline is not worth a stuck shard."""
+        period = max(2.0, self._session_timeout_ms / 1000.0)
+        while not self._closed:
+            await asyncio.sleep(period)
+            if self._zk is not None and not self._closed:
+                self._enqueue(self._handle_active)
+                self._enqueue(self._handle_cluster_state)
 
     async def _setup_client(self) -> None:
         self._zk = ZkClient(self._conn_str,
@@ -171,6 +189,9 @@ class ZkMgr:
 
     async def close(self) -> None:
         self._closed = True
+        if getattr(self, "_resync_task", None) is not None:
+            self._resync_task.cancel()
+            self._resync_task = None
         if self._worker is not None:
             await self._tasks.put(None)
             try:
@@ -239,6 +260,17 @@ class ZkMgr:
                 peer.data = {}
         return peers
 
+    def _retry_handler(self, coro_fn: Callable, delay_s: float = 0.25
+                       ) -> None:
+        """A one-shot watch was just CONSUMED and re-arming it failed —
+        without a retry this peer would be permanently blind to that
+        node (no watch armed, nothing else re-arms it until session
+        expiry, which may never come on a healthy connection)."""
+        if self._closed:
+            return
+        asyncio.get_running_loop().call_later(
+            delay_s, self._enqueue, coro_fn)
+
     async def _handle_active(self) -> None:
         """Re-list + re-watch the election dir; emit activeChange only when
         the membership actually changed (ref handleActive :307-386)."""
@@ -248,7 +280,9 @@ class ZkMgr:
             children, _ = await self._zk.get_children(
                 self._election_path, watch=self._on_children_event)
         except jute.ZkError as exc:
-            self.log.warn("election re-list failed", err=exc)
+            self.log.warn("election re-list failed; retrying until the "
+                          "watch is re-armed", err=exc)
+            self._retry_handler(self._handle_active)
             return
         peers = parse_and_unique_actives(children)
         if id_lists_equal(peers, self._active):
@@ -287,7 +321,9 @@ class ZkMgr:
         try:
             state = await self._read_state_and_watch()
         except jute.ZkError as exc:
-            self.log.warn("state re-read failed", err=exc)
+            self.log.warn("state re-read failed; retrying until the "
+                          "watch is re-armed", err=exc)
+            self._retry_handler(self._handle_cluster_state)
             return
         if state is None:
             return

@@ -206,3 +206,80 @@ def test_session_expiry_rebuild():
             await m1.close()
             await srv.stop()
     run(go())
+
+
+# ------------------------------------------------- watch-loss regressions
+def test_transient_relist_failure_does_not_lose_the_watch():
+    """A one-shot watch fires, the re-list RPC fails transiently
+    (connection blip): the handler must retry until the watch is
+    re-armed — returning without one leaves the peer permanently blind
+    to membership changes (the takeover-never-happens failure shape)."""
+    async def go():
+        srv = ZkServer()
+        await srv.start()
+        m1 = mk_mgr(srv, "10.0.0.1", timeout=2000)
+        changes = []
+        m1.on("activeChange", lambda a: changes.append([p["id"]
+                                                       for p in a]))
+        await m1.init()
+        m2 = mk_mgr(srv, "10.0.0.2", timeout=2000)
+        await m2.init()
+        try:
+            await wait_for(lambda: any("10.0.0.2:5432:12345" in c
+                                       for c in changes),
+                           what="m2 join observed")
+            # make the NEXT re-list fail once, then work again
+            orig = m1._zk.get_children
+            state = {"fails": 1}
+
+            async def flaky(path, watch=None):
+                if state["fails"] > 0:
+                    state["fails"] -= 1
+                    raise jute.ZkError(jute.ZCONNECTIONLOSS, path)
+                return await orig(path, watch=watch)
+
+            m1._zk.get_children = flaky
+            before = len(changes)
+            await m2.close()     # fires m1's watch; re-list fails once
+            await wait_for(lambda: len(changes) > before, timeout=10,
+                           what="departure observed despite the blip")
+            assert "10.0.0.2:5432:12345" not in changes[-1]
+            assert state["fails"] == 0
+        finally:
+            await m1.close()
+            await srv.stop()
+    run(go())
+
+
+def test_resync_heals_a_fully_lost_watch():
+    """Belt and braces: even if a watch notification is lost outright
+    (simulated by discarding it), the low-frequency resync loop must
+    surface the membership change within ~a session timeout."""
+    async def go():
+        srv = ZkServer()
+        await srv.start()
+        m1 = mk_mgr(srv, "10.0.0.1", timeout=2000)
+        changes = []
+        m1.on("activeChange", lambda a: changes.append([p["id"]
+                                                       for p in a]))
+        await m1.init()
+        m2 = mk_mgr(srv, "10.0.0.2", timeout=2000)
+        await m2.init()
+        try:
+            await wait_for(lambda: any("10.0.0.2:5432:12345" in c
+                                       for c in changes),
+                           what="m2 join observed")
+            # lose the watch outright: drop the client-side registration
+            # so the server's notification finds nothing to dispatch and
+            # nothing re-arms
+            m1._zk._child_watches.clear()
+            before = len(changes)
+            await m2.close()
+            # the resync loop (period = session timeout) must notice
+            await wait_for(lambda: len(changes) > before, timeout=10,
+                           what="departure observed via resync")
+            assert "10.0.0.2:5432:12345" not in changes[-1]
+        finally:
+            await m1.close()
+            await srv.stop()
+    run(go())
