@@ -248,6 +248,17 @@ async def run_rank(rank: int, steps: int, warmup: int, base_dir: str,
                          n_peers=3, shard_name="%d.bench" % (rank + 1),
                          engine=engine,
                          session_timeout_ms=session_timeout_ms)
+    # SIGTERM (e.g. `timeout`-bounded runs) must unwind through the
+    # finally below — a hard exit would orphan the whole cluster, and a
+    # leaked db squatting a port poisons later runs on the same host
+    loop = asyncio.get_running_loop()
+    me = asyncio.current_task()
+    import signal as _signal
+    for _sig in (_signal.SIGTERM, _signal.SIGINT):
+        try:
+            loop.add_signal_handler(_sig, me.cancel)
+        except (NotImplementedError, RuntimeError):
+            pass
     results = []
     total_lost = 0
     total_checked = 0
