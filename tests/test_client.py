@@ -147,3 +147,55 @@ def test_client_sees_live_failover(tmp_path):
                 await mc.close()
             c.stop()
     run(go())
+
+
+def test_watch_poke_during_read_is_not_swallowed(tmp_path):
+    """Regression (advisor finding): a watch notification dispatched
+    BETWEEN re-arming the watch and waiting on the poke event must not
+    be swallowed.  The client must clear the poke BEFORE the
+    read+rewatch; a clear placed after the read would eat an event set
+    mid-read, block forever on a consumed watch, and serve stale
+    topology until session expiry."""
+    async def go():
+        srv = ZkServer()
+        await srv.start()
+        zk = ZkClient(srv.conn_str, session_timeout_ms=30000)
+        await zk.connect()
+        await zk.mkdirp("/manatee/1.poke")
+        await zk.create("/manatee/1.poke/state",
+                        json.dumps(state(1, 1, 2)).encode())
+
+        cli = ManateeClient(srv.conn_str, "1.poke")
+        reads = {"n": 0}
+        orig = cli._read_and_watch
+
+        async def racing_read():
+            await orig()
+            # simulate the io-loop dispatching a buffered notification
+            # right after the watch was re-armed, before wait() runs
+            if reads["n"] == 0:
+                cli._poke.set()
+            reads["n"] += 1
+
+        cli._read_and_watch = racing_read
+        await cli.start()
+        try:
+            # the mid-read poke must cause ANOTHER read cycle (the old
+            # ordering swallowed it and blocked in wait() forever)
+            deadline = asyncio.get_running_loop().time() + 5
+            while reads["n"] < 2:
+                assert asyncio.get_running_loop().time() < deadline, \
+                    "poke set during read was swallowed; client blind"
+                await asyncio.sleep(0.02)
+            # and a real state change still comes through afterwards
+            await zk.set_data("/manatee/1.poke/state",
+                              json.dumps(state(2, 2, 1)).encode())
+            deadline = asyncio.get_running_loop().time() + 5
+            while (cli.topology or {}).get("generation") != 2:
+                assert asyncio.get_running_loop().time() < deadline
+                await asyncio.sleep(0.02)
+        finally:
+            await cli.close()
+            await zk.close()
+            await srv.stop()
+    run(go())
