@@ -787,3 +787,123 @@ def test_post_restore_fixup_purges_peer_local_files(tmp_path):
     assert left == ["waldb_ident.json"], left
     assert not os.path.exists(
         os.path.join(tmp_path / "store" / "live", "waldb.log"))
+
+
+def test_sync_ack_pins_wal_retention_across_disconnect(tmp_path):
+    """Regression (advisor finding): the NAMED sync standby's last acked
+    LSN must pin the WAL retention floor across a brief disconnect —
+    otherwise a sync that blips while the primary appends past the keep
+    window gets 'wal-gone' on reconnect and is forced into a full
+    restore (read-only storm under load)."""
+    async def go():
+        prim = Node(tmp_path, "prim")
+        prim.init()
+        # tiny windows so a burst of appends would normally recycle
+        # everything the sync still needs
+        extra = {"checkpoint_wal_bytes": "4096",
+                 "wal_keep_bytes": "2048",
+                 "wal_segment_bytes": "8192"}
+        prim.write_conf(role="primary", sync_name="sync", extra=extra)
+        prim.start()
+        sync = Node(tmp_path, "sync")
+        try:
+            pcli = prim.client()
+            # replicate + ack a first batch
+            import shutil
+            shutil.copytree(os.path.join(prim.data_dir),
+                            sync.data_dir, dirs_exist_ok=True)
+            for name in ("waldb.conf", "waldb.pid"):
+                try:
+                    os.unlink(os.path.join(sync.data_dir, name))
+                except FileNotFoundError:
+                    pass
+            sync.write_conf(role="standby",
+                            upstream="127.0.0.1:%d" % prim.port,
+                            extra={"name": "'sync'"})
+            sync.start()
+            await pcli.put("base", "v")
+
+            async def caught_up():
+                st = await pcli.status()
+                return any(r["application_name"] == "sync"
+                           and r["write_lsn"] == st["current_lsn"]
+                           for r in st["replication"])
+            await wait_async(caught_up, what="sync caught up")
+            st = await pcli.status()
+            acked_at = st["current_lsn"]
+
+            # sync blips; the primary keeps APPENDING (each gated put
+            # appends before blocking on the ack, like a client retry
+            # storm) far past the keep window, with checkpoints running
+            sync.kill9()
+            for i in range(40):
+                try:
+                    await pcli.put("burst%d" % i, "x" * 512,
+                                   timeout_s=0.15)
+                except WaldbError:
+                    pass
+            await asyncio.sleep(0.5)   # let checkpoints recycle
+
+            st = await pcli.status()
+            from manatee_amd.common.lsn import parse
+            assert parse(st["wal_start_lsn"]) <= parse(acked_at), \
+                "retention floor did not pin the disconnected sync's " \
+                "acked LSN (start %s > acked %s)" \
+                % (st["wal_start_lsn"], acked_at)
+
+            # the sync reconnects and STREAMS (no wal-gone → no restore)
+            sync.start()
+            scli = sync.client()
+
+            async def streaming():
+                st2 = await scli.status()
+                return st2["upstream_status"] == "streaming"
+            await wait_async(streaming, what="sync resumed streaming")
+            await scli.close()
+            await pcli.close()
+        finally:
+            prim.stop()
+            sync.stop()
+    run(go())
+
+
+def test_checkpoint_fsyncs_wal_before_publishing(tmp_path):
+    """Regression (advisor finding): the checkpoint must fsync the WAL
+    BEFORE capturing/publishing its LSN — a checkpoint.lsn beyond the
+    durable WAL end would let post-crash recovery reuse LSNs below
+    already-streamed positions on the same timeline."""
+    async def go():
+        from manatee_amd.common.logging import null_logger
+        from manatee_amd.db.waldb.server import WaldbServer, CKPT_NAME
+
+        data = str(tmp_path / "n1")
+        init_data_dir(data)
+        confparser.write(os.path.join(data, "waldb.conf"), {
+            "role": "primary", "listen_ip": "127.0.0.1", "port": "0",
+            "name": "n1"})
+        srv = WaldbServer(data, null_logger())
+        await srv.start()
+        try:
+            for i in range(10):
+                await srv._do_write({"op": "put", "k": "k%d" % i,
+                                     "v": "v"})
+            events = []
+            orig_fsync = srv.wal.fsync
+
+            def spying_fsync():
+                events.append(srv.wal.end)
+                return orig_fsync()
+
+            srv.wal.fsync = spying_fsync
+            await srv._checkpoint()
+            with open(os.path.join(data, CKPT_NAME)) as f:
+                ckpt = json.load(f)
+            assert events, "checkpoint did not fsync the WAL first"
+            assert events[0] >= ckpt["lsn"], \
+                "checkpoint LSN published beyond the fsynced WAL end"
+        finally:
+            if srv._server is not None:
+                srv._server.close()
+            if srv._flusher is not None:
+                srv._flusher.cancel()
+    run(go())
