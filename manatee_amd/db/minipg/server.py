@@ -274,16 +274,67 @@ class MinipgServer(WaldbServer):
                                writer: asyncio.StreamWriter) -> None:
         """One simple-Query cycle: possibly multiple statements, one
         ReadyForQuery at the end; an error aborts the rest (as the real
-        backend does)."""
+        backend does).  GROUP COMMIT: every write in the statement list
+        is appended immediately and the whole cycle waits once for the
+        highest LSN's sync ack, so a multi-statement INSERT batch pays
+        one replication round trip, not one per row."""
+        staged: List[bytes] = []
+        max_lsn = None
+        err: Optional[PgSqlError] = None
         for stmt in split_statements(sql) or [""]:
             try:
-                for chunk in await self._execute(stmt):
-                    writer.write(chunk)
+                chunks, lsn = await self._execute_staged(stmt)
             except PgSqlError as exc:
-                writer.write(self._error(exc.code, exc.msg))
+                err = exc
                 break
+            staged.extend(chunks)
+            if lsn is not None:
+                max_lsn = lsn
+        if max_lsn is not None:
+            res = await self._await_commit(max_lsn)
+            if not res.get("ok"):
+                # gate broke while waiting (e.g. read-only flipped):
+                # nothing in this cycle may be acknowledged
+                staged = []
+                err = err or PgSqlError("25006", res.get("error",
+                                                         "commit failed"))
+        for chunk in staged:
+            writer.write(chunk)
+        if err is not None:
+            writer.write(self._error(err.code, err.msg))
         writer.write(self._ready())
         await writer.drain()
+
+    async def _execute_staged(self, stmt: str
+                              ) -> Tuple[List[bytes], Optional[int]]:
+        """Writes append-without-waiting (their acks are withheld until
+        the cycle's single commit gate); everything else executes
+        directly."""
+        m = self._RE_INSERT.search(stmt)
+        if m:
+            lsn = self._append_sql(
+                {"op": "put", "k": sql_unquote(m.group(1)),
+                 "v": sql_unquote(m.group(2))}, "INSERT")
+            return [self._complete("INSERT 0 1")], lsn
+        m = self._RE_DELETE.search(stmt)
+        if m:
+            k = sql_unquote(m.group(1))
+            if k not in self.kv:
+                return [self._complete("DELETE 0")], None
+            lsn = self._append_sql({"op": "del", "k": k}, "DELETE")
+            return [self._complete("DELETE 1")], lsn
+        return await self._execute(stmt), None
+
+    def _append_sql(self, op: dict, verb: str) -> int:
+        lsn, err = self._append_write(op)
+        if err is not None:
+            msg = err.get("error", "")
+            if "read-only" in msg:
+                raise PgSqlError(
+                    "25006",
+                    "cannot execute %s in a read-only transaction" % verb)
+            raise PgSqlError("XX000", msg or "write failed")
+        return lsn
 
     # ------------------------------------------------------------ SQL layer
     _RE_INSERT = re.compile(
@@ -360,10 +411,6 @@ class MinipgServer(WaldbServer):
         if "pg_stat_replication" in low:
             return self._stat_replication()
 
-        m = self._RE_INSERT.search(stmt)
-        if m:
-            return await self._sql_insert(sql_unquote(m.group(1)),
-                                          sql_unquote(m.group(2)))
         m = self._RE_SELECT_V.search(stmt)
         if m:
             k = sql_unquote(m.group(1))
@@ -379,30 +426,8 @@ class MinipgServer(WaldbServer):
                 if prefix else len(self.kv)
             return [self._row_desc(["count"]), self._data_row([str(n)]),
                     self._complete("SELECT 1")]
-        m = self._RE_DELETE.search(stmt)
-        if m:
-            k = sql_unquote(m.group(1))
-            if k not in self.kv:
-                return [self._complete("DELETE 0")]
-            await self._sql_write({"op": "del", "k": k}, "DELETE")
-            return [self._complete("DELETE 1")]
-
         raise PgSqlError("42601", 'syntax error at or near "%s"'
                          % stmt.split()[0][:40])
-
-    async def _sql_insert(self, k: str, v: str) -> List[bytes]:
-        await self._sql_write({"op": "put", "k": k, "v": v}, "INSERT")
-        return [self._complete("INSERT 0 1")]
-
-    async def _sql_write(self, op: dict, verb: str) -> None:
-        res = await self._do_write(op)
-        if not res.get("ok"):
-            err = res.get("error", "")
-            if "read-only" in err:
-                raise PgSqlError(
-                    "25006",
-                    "cannot execute %s in a read-only transaction" % verb)
-            raise PgSqlError("XX000", err or "write failed")
 
     def _stat_replication(self) -> List[bytes]:
         w = self.lsn_word
