@@ -73,14 +73,15 @@ def test_version_info_cases(tmp_path):
         get_version_info(data, dconf, versions, "12")
 
 
-def mk_engine(tmp_path, major, versions=None):
+def mk_engine(tmp_path, major, versions=None, extra_cfg=None):
     versions = versions or {"9.6": "9.6.3", "12": "12.0"}
     data = str(tmp_path / "store" / "data")
     os.makedirs(data, exist_ok=True)
+    cfg = {"versions": versions, "defaultVersion": major,
+           "pgBaseDir": str(tmp_path / "pg")}
+    cfg.update(extra_cfg or {})
     return PostgresEngine(
-        data, "10.0.0.1", 5432, "10.0.0.1:5432:5434",
-        cfg={"versions": versions, "defaultVersion": major,
-             "pgBaseDir": str(tmp_path / "pg")})
+        data, "10.0.0.1", 5432, "10.0.0.1:5432:5434", cfg=cfg)
 
 
 def test_conf_generation_pg96_standby_uses_recovery_conf(tmp_path):
@@ -206,3 +207,20 @@ def test_pgwire_client_roundtrip():
         server.close()
         await server.wait_closed()
     asyncio.run(asyncio.wait_for(go(), 30))
+
+
+def test_full_page_writes_default_is_safe(tmp_path):
+    """Regression (advisor finding): the reference's template turns
+    full_page_writes OFF assuming ZFS (copy-on-write, no torn pages);
+    on a plain filesystem that risks unrecoverable torn-page corruption
+    after power loss.  The safe value must be the default, relaxed only
+    when the configured store is copy-on-write."""
+    eng = mk_engine(tmp_path, "12")
+    eng.write_conf("primary")
+    conf = confparser.read(eng._conf_path())
+    assert conf["full_page_writes"] == "on"      # safe default (DirStore)
+
+    eng2 = mk_engine(tmp_path, "12", extra_cfg={"storeIsCow": True})
+    eng2.write_conf("primary")
+    conf = confparser.read(eng2._conf_path())
+    assert conf["full_page_writes"] == "off"     # ZFS: reference value
