@@ -372,34 +372,66 @@ async def cmd_zk_active(ns) -> int:
 
 
 async def cmd_history(ns) -> int:
+    """ref do_history bin/manatee-adm: table of TIME/G#/MODE/FRZ/
+    PRIMARY/SYNC/ASYNC/DEPOSED (zoneId[:8]), -v adds SUMMARY (the
+    legal-transition annotation), -j emits {zkSeq, time, state} JSON
+    lines, -s sorts by zkSeq (default) or time."""
+    if getattr(ns, "sort", "zkSeq") not in ("zkSeq", "time"):
+        raise UsageError('-s / --sort must be one of "zkSeq" or "time"')
     shard = _need(ns, "shard", "SHARD", "-s/--shard")
+
+    def abbr(ident) -> str:
+        if not ident:
+            return "-"
+        return (ident.get("zoneId") or ident.get("id") or "-")[:8]
 
     async def go(zk):
         entries = await adm.get_history(zk, shard)
         annotated = adm.annotate_history(entries)
+        if ns.sort == "time":
+            annotated.sort(key=lambda e: e.get("time") or 0)
         if ns.json:
             for e in annotated:
-                print(json.dumps(e, sort_keys=True))
+                print(json.dumps(
+                    {"zkSeq": e["zkSeq"],
+                     "time": (st.iso8601(e["time"] / 1000.0)
+                              if e.get("time") else None),
+                     "state": e["state"]}))
             return 0
+        cols = [("TIME", 24, "l"), ("G#", 2, "r"), ("MODE", 5, "l"),
+                ("FRZ", 3, "l"), ("PRIMARY", 8, "l"), ("SYNC", 8, "l"),
+                ("ASYNC", 8, "l"), ("DEPOSED", 8, "l")]
+        if ns.verbose:
+            cols.append(("SUMMARY", 0, "l"))
+
+        def emit(row):
+            cells = []
+            for (label, width, align), val in zip(cols, row):
+                text = str(val)
+                if width:
+                    text = (text.rjust(width) if align == "r"
+                            else text.ljust(width))
+                cells.append(text)
+            print(" ".join(cells))
+
+        emit([c[0] for c in cols])
         for e in annotated:
             s = e["state"]
-            when = st.iso8601(e["time"] / 1000.0) if e["time"] else "-"
-            line = "%-4s %-24s gen %-3s primary %-8s sync %-8s" % (
-                e["zkSeq"], when, s.get("generation"),
-                (s.get("primary") or {}).get("id", "-")[:8],
-                ((s.get("sync") or {}).get("id", "-") or "-")[:8])
-            asyncs = ",".join(a["id"][:8] for a in s.get("async") or [])
-            deposed = ",".join(d["id"][:8] for d in s.get("deposed") or [])
-            line += " async [%s]" % asyncs
-            if deposed:
-                line += " deposed [%s]" % deposed
-            if s.get("freeze"):
-                line += " FROZEN"
-            print(line)
-            for note in e.get("notes", []):
-                print("         %s" % note)
-            for viol in e.get("violations", []):
-                print("         VIOLATION: %s" % viol)
+            when = st.iso8601(e["time"] / 1000.0) if e.get("time") else "-"
+            asyncs = ",".join(abbr(a) for a in s.get("async") or []) or "-"
+            deposed = ",".join(abbr(d)
+                               for d in s.get("deposed") or []) or "-"
+            row = [when, s.get("generation", "-"),
+                   "singl" if s.get("oneNodeWriteMode") else "multi",
+                   "frz" if s.get("freeze") else "-",
+                   abbr(s.get("primary")), abbr(s.get("sync")),
+                   asyncs, deposed]
+            if ns.verbose:
+                summary = "; ".join(
+                    list(e.get("notes") or []) +
+                    ["VIOLATION: " + v for v in e.get("violations") or []])
+                row.append(summary)
+            emit(row)
         return 0
     return await _with_zk(ns, go)
 
@@ -744,6 +776,10 @@ def _mk_parser() -> argparse.ArgumentParser:
     sp = add("history", cmd_history,
              help="cluster state history with transition checks")
     sp.add_argument("-j", "--json", action="store_true")
+    sp.add_argument("-v", "--verbose", action="store_true",
+                    help="add the SUMMARY column (transition annotation)")
+    sp.add_argument("-S", "--sort", default="zkSeq",
+                    help='"zkSeq" (default) or "time"')
 
     sp = add("freeze", cmd_freeze, help="freeze cluster transitions")
     sp.add_argument("-r", "--reason")

@@ -262,7 +262,18 @@ def test_adm_cli_live_against_cluster(cluster_dir):
 
             r = cli("history")
             assert r.returncode == 0
-            assert "cluster setup for normal" in r.stdout
+            # reference-shaped table (TIME/G#/MODE/... with zoneId abbrs)
+            assert r.stdout.splitlines()[0].startswith(
+                "TIME                     G# MODE  FRZ PRIMARY")
+            assert " multi " in r.stdout
+            r = cli("history", "-v")
+            assert r.returncode == 0
+            assert "cluster setup for normal" in r.stdout   # SUMMARY col
+            r = cli("history", "-j")
+            assert r.returncode == 0
+            ev = json.loads(r.stdout.splitlines()[0])
+            assert set(ev) == {"zkSeq", "time", "state"}
+            assert ev["state"]["generation"] == 1
 
             r = cli("status")
             assert r.returncode == 0
