@@ -347,16 +347,17 @@ async def cmd_zk_state(ns) -> int:
     fx = det.fixture_path()
     if fx:
         cd = det.load_fixture(fx)
-        print(json.dumps(cd.state, indent=2, sort_keys=True))
+        print(json.dumps(cd.state, separators=(",", ":")))
         return 0
     shard = _need(ns, "shard", "SHARD", "-s/--shard")
 
     async def go(zk):
         state, version = await adm.get_state(zk, shard)
         if state is None:
-            return _fail("no cluster state for shard %r" % shard)
-        state["_zkVersion"] = version
-        print(json.dumps(state, indent=2, sort_keys=True))
+            return _fail("No state exists for shard " + shard)
+        # the reference prints JSON.stringify(state): compact, key order
+        # preserved, no extra fields (ref do_zk_state / zkState)
+        print(json.dumps(state, separators=(",", ":")))
         return 0
     return await _with_zk(ns, go)
 
