@@ -68,9 +68,27 @@ class PgClient:
 
     # ------------------------------------------------------------ lifecycle
     async def connect(self) -> None:
-        self._reader, self._writer = await asyncio.wait_for(
-            dial.open_connection(self.host, self.port),
-            self.connect_timeout_s)
+        """TCP connect + startup/auth handshake, ALL bounded by
+        connect_timeout_s: a backend that accepts the connection but
+        never answers the startup packet (dying process, half-open
+        socket after a partition) must not hang the caller forever."""
+        try:
+            await asyncio.wait_for(self._connect(),
+                                   self.connect_timeout_s)
+        except BaseException:
+            writer = self._writer
+            self._writer = None
+            self._reader = None
+            if writer is not None:
+                try:
+                    writer.close()
+                except Exception:
+                    pass
+            raise
+
+    async def _connect(self) -> None:
+        self._reader, self._writer = await dial.open_connection(
+            self.host, self.port)
         params = ("user\x00%s\x00database\x00%s\x00\x00"
                   % (self.user, self.database)).encode("utf-8")
         body = _I32.pack(PROTOCOL_VERSION) + params

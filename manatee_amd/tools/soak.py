@@ -145,12 +145,24 @@ ACTIONS = ["kill_primary", "kill_sync", "kill_async",
 
 
 async def soak(minutes: float, seed: int, workdir: str,
-               cycles: Optional[int] = None) -> dict:
+               cycles: Optional[int] = None,
+               engine: str = "waldb") -> dict:
     rng = random.Random(seed)
-    c = DevCluster(workdir, n_peers=3, shard_name="1.soak", proxied=True)
+    c = DevCluster(workdir, n_peers=3, shard_name="1.soak", proxied=True,
+                   engine=engine)
     stats = {"cycles": 0, "kills": {}, "lost": 0, "acked": 0,
              "max_failover_s": 0.0, "failures": []}
     writer = SoakWriter(c)
+    # SIGTERM (timeout-bounded runs) must unwind through the finally so
+    # the cluster is torn down, not orphaned
+    import signal as _signal
+    loop = asyncio.get_running_loop()
+    me = asyncio.current_task()
+    for _sig in (_signal.SIGTERM, _signal.SIGINT):
+        try:
+            loop.add_signal_handler(_sig, me.cancel)
+        except (NotImplementedError, RuntimeError):
+            pass
     try:
         await c.start()
         await c.wait_cluster(
@@ -287,17 +299,20 @@ def main(argv=None) -> int:
     ap.add_argument("--cycles", type=int, default=None,
                     help="stop after N fault cycles (within the budget)")
     ap.add_argument("--seed", type=int, default=0)
+    ap.add_argument("--engine", choices=("waldb", "postgres"),
+                    default="waldb")
     ap.add_argument("-d", "--dir", default=None)
     ns = ap.parse_args(argv)
     workdir = ns.dir or tempfile.mkdtemp(prefix="manatee-soak-")
     try:
         stats = asyncio.run(soak(ns.minutes, ns.seed, workdir,
-                                 cycles=ns.cycles))
+                                 cycles=ns.cycles, engine=ns.engine))
     finally:
         if ns.dir is None:
             shutil.rmtree(workdir, ignore_errors=True)
     ok = not stats["lost"] and not stats["failures"]
     stats["ok"] = ok
+    stats["engine"] = ns.engine
     print(json.dumps(stats))
     return 0 if ok else 1
 

@@ -203,3 +203,45 @@ def test_asymmetric_partition_acks_lost_one_way(tmp_path):
         finally:
             c.stop()
     run(go())
+
+
+def test_primary_cut_from_zk_postgres_engine(tmp_path):
+    """The same classic split-brain shape through the engine=postgres
+    path: minipg's replication and the libpq probes all dial through
+    the per-link proxies; the deposed primary must refuse acks and no
+    acked write may be lost."""
+    async def go():
+        c = DevCluster(str(tmp_path / "c"), n_peers=3,
+                       shard_name="1.pgpartzk", session_timeout_ms=2000,
+                       proxied=True, engine="postgres",
+                       run_snapshotter=False)
+        try:
+            await c.start()
+            await c.wait_cluster(
+                lambda s: s.get("sync") and len(s.get("async", [])) == 1,
+                timeout_s=120, what="formation (postgres, proxied)")
+            await c.wait_writable(timeout_s=120)
+            w = bench.Writer(c)
+            w.start()
+            while w.seq < 25:
+                await asyncio.sleep(0.05)
+
+            s0 = await c.cluster_state()
+            prim = c.peer_by_id(s0["primary"]["id"])
+            c.partition_zk(prim)
+            s1 = await c.wait_cluster(
+                lambda s: s["generation"] > s0["generation"],
+                timeout_s=30, what="takeover under zk partition (pg)")
+            await c.wait_writable(timeout_s=30)
+            await _put_must_fail(prim, "pg-split-brain-probe")
+            v = await bench.verify_no_loss(c, s1, w)
+            assert v["lost"] == 0, v
+            c.heal_zk(prim)
+            await c.wait_cluster(
+                lambda s: any(d["id"] == prim.id
+                              for d in s.get("deposed", [])),
+                timeout_s=30, what="old primary deposed (pg)")
+            await w.stop()
+        finally:
+            c.stop()
+    run(go())
