@@ -323,7 +323,16 @@ class FakeZfs:
         if parent and not self.exists(parent):
             raise ZfsError("cannot receive: parent '%s' does not exist"
                            % parent)
-        header = sys.stdin.buffer.readline()
+        # read the header byte-by-byte from the RAW fd: a buffered
+        # readline() would read ahead past the newline and swallow the
+        # start of the tar stream before tar inherits the fd
+        fd = sys.stdin.buffer.fileno()
+        header = bytearray()
+        while not header.endswith(b"\n"):
+            b = os.read(fd, 1)
+            if not b:
+                break
+            header += b
         if not header.startswith(b"FAKEZFS1 "):
             raise ZfsError("invalid stream (bad magic)")
         snap_name = header.split(b" ", 1)[1].strip().decode()
