@@ -72,10 +72,15 @@ class BackupSender:
     (ref lib/backupSender.js:154-242)."""
 
     def __init__(self, store: SnapshotStore, queue: BackupQueue,
-                 log: Optional[Logger] = None):
+                 log: Optional[Logger] = None,
+                 stall_timeout_s: float = 120.0):
         self.store = store
         self.queue = queue
         self.log = (log or null_logger()).child(component="BackupSender")
+        # a receiver that stops reading (partitioned mid-restore, dead
+        # process with the TCP half open) must fail the job, not wedge
+        # the SERIAL sender queue forever
+        self.stall_timeout_s = stall_timeout_s
         self._task: Optional[asyncio.Task] = None
 
     def start(self) -> None:
@@ -121,10 +126,16 @@ class BackupSender:
             stream = await self.store.send(job.snapshot)
             async for chunk in stream:
                 writer.write(chunk)
-                await writer.drain()
+                try:
+                    await asyncio.wait_for(writer.drain(),
+                                           self.stall_timeout_s)
+                except asyncio.TimeoutError:
+                    raise RuntimeError(
+                        "receiver stalled: no progress in %.0fs"
+                        % self.stall_timeout_s)
                 job.completed += len(chunk)
             writer.write_eof()
-            await writer.drain()
+            await asyncio.wait_for(writer.drain(), self.stall_timeout_s)
             job.done = True
             self.log.info("backup job complete", jobid=job.uuid,
                           bytes=job.completed)
