@@ -209,10 +209,14 @@ class MinipgServer(WaldbServer):
     async def _serve_libpq(self, first: bytes,
                            reader: asyncio.StreamReader,
                            writer: asyncio.StreamWriter) -> None:
+        MAX_MSG = 16 * 1024 * 1024
         try:
             while True:     # startup negotiation (SSL probe then startup)
                 rest = await reader.readexactly(3)
                 (length,) = struct.unpack(">I", first + rest)
+                if length < 4 or length - 4 > MAX_MSG:
+                    writer.close()
+                    return
                 payload = await reader.readexactly(length - 4)
                 (code,) = struct.unpack_from(">i", payload)
                 if code in (SSL_REQUEST, GSSENC_REQUEST):
@@ -247,6 +251,11 @@ class MinipgServer(WaldbServer):
                 hdr = await reader.readexactly(5)
                 t = hdr[:1]
                 (ln,) = struct.unpack(">I", hdr[1:])
+                if ln < 4 or ln - 4 > MAX_MSG:
+                    writer.write(self._error(
+                        "08P01", "invalid message length %d" % ln))
+                    await writer.drain()
+                    return
                 payload = await reader.readexactly(ln - 4)
                 if t == b"X":
                     return
