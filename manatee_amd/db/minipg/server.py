@@ -151,6 +151,10 @@ class MinipgServer(WaldbServer):
                 "wal_keep_bytes", str(self.wal_keep_bytes))))
             self.wal.segment_bytes = int(s(raw.get(
                 "wal_segment_bytes", str(self.wal.segment_bytes))))
+            # PG expresses wal_sender_timeout in milliseconds
+            self.wal_sender_timeout_s = float(s(raw.get(
+                "wal_sender_timeout",
+                str(int(self.wal_sender_timeout_s * 1000))))) / 1000.0
         except ValueError:
             pass
 
