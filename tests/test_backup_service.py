@@ -87,7 +87,9 @@ def test_stalled_receiver_fails_job_without_wedging_queue(tmp_path):
     timeout, and the NEXT job must still be served — one dead receiver
     may not block every future bootstrap from this peer."""
     async def go():
-        st = await _store_with_data(tmp_path, nbytes=4 * 1024 * 1024)
+        # the payload must overflow every OS socket buffer on the way,
+        # or the 'stalled' receiver silently absorbs the whole stream
+        st = await _store_with_data(tmp_path, nbytes=64 * 1024 * 1024)
         q = BackupQueue()
         sender = BackupSender(st, q, log=null_logger(),
                               stall_timeout_s=1.0)
@@ -96,7 +98,12 @@ def test_stalled_receiver_fails_job_without_wedging_queue(tmp_path):
         async def black_hole(reader, writer):
             await asyncio.sleep(3600)   # accept, never read
 
-        stalled = await asyncio.start_server(black_hole, "127.0.0.1", 0)
+        import socket as socketmod
+        bh_sock = socketmod.socket()
+        bh_sock.setsockopt(socketmod.SOL_SOCKET,
+                           socketmod.SO_RCVBUF, 4096)
+        bh_sock.bind(("127.0.0.1", 0))
+        stalled = await asyncio.start_server(black_hole, sock=bh_sock)
         sport = stalled.sockets[0].getsockname()[1]
 
         good = bytearray()
