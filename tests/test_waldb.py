@@ -907,3 +907,72 @@ def test_checkpoint_fsyncs_wal_before_publishing(tmp_path):
             if srv._flusher is not None:
                 srv._flusher.cancel()
     run(go())
+
+
+def test_group_commit_pipeline_order_and_gate_break(tmp_path):
+    """Group commit: a pipelined burst is answered in request order with
+    every record acknowledged after ONE sync-ack wait; and when the
+    commit gate breaks mid-burst (read-only flipped while commits are
+    pending), the pending writes are answered with errors, in order,
+    without wedging the connection."""
+    prim = Node(tmp_path, "prim")
+    sync = Node(tmp_path, "sync")
+    prim.init()
+    prim.write_conf(role="primary", sync_name="sync")
+    prim.start()
+    try:
+        import shutil
+        shutil.copytree(prim.data_dir, sync.data_dir,
+                        ignore=shutil.ignore_patterns("waldb.pid",
+                                                      "waldb.conf"))
+        sync.write_conf(role="standby",
+                        upstream="127.0.0.1:%d" % prim.port)
+        sync.start()
+
+        async def burst_ok():
+            c = prim.client()
+            sc = sync.client()
+
+            async def streaming():
+                st = await sc.status()
+                return st["upstream_status"] == "streaming"
+            await wait_async(streaming, what="sync streaming")
+            resps = await c.pipeline(
+                [{"q": "put", "k": "gc%d" % i, "v": i}
+                 for i in range(60)], timeout_s=10.0)
+            assert all(r["ok"] for r in resps)
+            # responses are in request order: LSNs strictly increase
+            from manatee_amd.common.lsn import parse
+            lsns = [parse(r["lsn"]) for r in resps]
+            assert lsns == sorted(lsns) and len(set(lsns)) == 60
+            assert await c.count(prefix="gc") == 60
+            await c.close()
+            await sc.close()
+        run(burst_ok())
+
+        # break the gate mid-burst: kill the sync, start a pipelined
+        # burst (appends immediately, commits pending), then flip
+        # read-only via conf reload — the pending burst must be answered
+        # with errors, not hang
+        sync.kill9()
+
+        async def burst_gate_break():
+            c = prim.client()
+            burst = asyncio.get_running_loop().create_task(
+                c.pipeline([{"q": "put", "k": "rb%d" % i, "v": i}
+                            for i in range(10)], timeout_s=20.0))
+            await asyncio.sleep(0.5)    # burst is appended + gated
+            assert not burst.done()
+            prim.write_conf(role="primary", sync_name="sync",
+                            read_only=True)
+            prim.sighup()
+            resps = await asyncio.wait_for(burst, 15.0)
+            assert all(not r["ok"] for r in resps)
+            assert all("read-only" in r["error"] for r in resps)
+            # the connection still serves queries afterwards
+            assert await c.count(prefix="gc") == 60
+            await c.close()
+        run(burst_gate_break())
+    finally:
+        prim.stop()
+        sync.stop()
