@@ -28,7 +28,17 @@ import struct
 import zlib
 from typing import Callable, Iterator, List, Optional, Tuple
 
-from ...native import codec as _native
+from ... import native as _native_pkg
+
+
+def _codec():
+    """The C++ codec module, or None.
+
+    Resolved through the package attribute at call time (not bound at
+    import) so a build that happens after this module was first imported
+    — e.g. the test fixture compiling the extension in a fresh checkout,
+    then ``importlib.reload(manatee_amd.native)`` — is picked up."""
+    return _native_pkg.codec
 
 _HDR = struct.Struct(">II")
 _SEG_RE = re.compile(r"^wal-([0-9a-f]{16})\.seg$")
@@ -145,8 +155,9 @@ class Wal:
     def append(self, payload: bytes) -> int:
         """Append one record; returns its commit LSN."""
         self._maybe_roll()
-        if _native is not None:
-            frame = _native.encode_frame(payload)
+        native = _codec()
+        if native is not None:
+            frame = native.encode_frame(payload)
         else:
             frame = _HDR.pack(len(payload), zlib.crc32(payload)) + payload
         os.pwrite(self._fd, frame, self.end - self._active_start)
@@ -281,8 +292,9 @@ class WalGone(Exception):
 def _scan(buf: bytes) -> Tuple[int, List[Tuple[int, int]]]:
     """Validate a record-stream prefix; returns (valid_bytes,
     [(payload_offset, payload_len), ...]).  Native-accelerated."""
-    if _native is not None:
-        valid, _count, offsets = _native.scan_records(buf, True)
+    native = _codec()
+    if native is not None:
+        valid, _count, offsets = native.scan_records(buf, True)
         return valid, offsets
     records: List[Tuple[int, int]] = []
     pos = 0
@@ -303,8 +315,9 @@ def parse_frames(data: bytes) -> Iterator[Tuple[int, bytes]]:
     """Split a raw replicated chunk into (frame_len, payload) records.
     The chunk always contains whole frames (senders send record-aligned).
     Native-accelerated when the codec extension is built."""
-    if _native is not None:
-        return iter(_native.parse_frames(data))
+    native = _codec()
+    if native is not None:
+        return iter(native.parse_frames(data))
     return _parse_frames_py(data)
 
 

@@ -1,7 +1,28 @@
+import subprocess
 import sys
 import os
 
-sys.path.insert(0, os.path.dirname(os.path.dirname(os.path.abspath(__file__))))
+_REPO = os.path.dirname(os.path.dirname(os.path.abspath(__file__)))
+sys.path.insert(0, _REPO)
+
+
+def _ensure_native_built():
+    """Compile the in-tree C++ extensions (codec, jutec) if they are not
+    built yet.  Runs at conftest import — before any test module imports
+    manatee_amd — so in a fresh checkout (the .so files are gitignored)
+    the whole suite still exercises the native paths instead of silently
+    binding the pure-Python fallbacks."""
+    native = os.path.join(_REPO, "manatee_amd", "native")
+    import glob
+    have = glob.glob(os.path.join(native, "_codec*.so")) and \
+        glob.glob(os.path.join(native, "_jutec*.so"))
+    if have:
+        return
+    subprocess.run([sys.executable, "setup.py", "build_ext", "--inplace"],
+                   cwd=native, check=True, capture_output=True, timeout=900)
+
+
+_ensure_native_built()
 
 
 def pytest_configure(config):

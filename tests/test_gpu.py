@@ -179,7 +179,7 @@ def test_native_codec_is_loaded_and_used():
     from manatee_amd.db.waldb import wal as walmod
     from manatee_amd.native import codec
     assert codec is not None, "native codec extension missing on the box"
-    assert walmod._native is codec
+    assert walmod._codec() is codec
     payload = b"x" * 1000
     frame = codec.encode_frame(payload)
     assert list(walmod.parse_frames(frame)) == [(len(frame), payload)]

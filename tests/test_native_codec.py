@@ -104,4 +104,4 @@ def _py_scan(buf):
 
 def test_wal_uses_native_when_built(codec):
     from manatee_amd.db.waldb import wal as walmod
-    assert walmod._native is not None
+    assert walmod._codec() is not None
