@@ -27,6 +27,7 @@ class WaldbClient:
         self.query_timeout_s = query_timeout_s
         self._reader: Optional[asyncio.StreamReader] = None
         self._writer: Optional[asyncio.StreamWriter] = None
+        self._rbuf = b""
         self._lock = asyncio.Lock()
 
     @classmethod
@@ -46,6 +47,24 @@ class WaldbClient:
             self._reader, self._writer = await asyncio.wait_for(
                 dial.open_connection(self.host, self.port),
                 self.connect_timeout_s)
+            self._rbuf = b""
+
+    async def _readline(self, timeout_s: float) -> bytes:
+        """Chunk-buffered line read: one read() syscall can serve a
+        whole pipelined response batch.  Returns b"" on EOF."""
+        buf = self._rbuf
+        nl = buf.find(b"\n")
+        while nl < 0:
+            chunk = await asyncio.wait_for(self._reader.read(65536),
+                                           timeout_s)
+            if not chunk:
+                self._rbuf = b""
+                return b""
+            scan_from = len(buf)        # no newline before the chunk
+            buf += chunk
+            nl = buf.find(b"\n", scan_from)
+        line, self._rbuf = buf[:nl + 1], buf[nl + 1:]
+        return line
 
     async def query(self, req: dict, timeout_s: Optional[float] = None
                     ) -> dict:
@@ -54,8 +73,7 @@ class WaldbClient:
             try:
                 self._writer.write((json.dumps(req) + "\n").encode())
                 await self._writer.drain()
-                line = await asyncio.wait_for(
-                    self._reader.readline(),
+                line = await self._readline(
                     timeout_s if timeout_s is not None
                     else self.query_timeout_s)
             except (ConnectionError, OSError, asyncio.TimeoutError) as exc:
@@ -78,6 +96,7 @@ class WaldbClient:
                 pass
         self._writer = None
         self._reader = None
+        self._rbuf = b""
 
     async def close(self) -> None:
         await self._teardown()
@@ -114,15 +133,14 @@ class WaldbClient:
                     json.dumps(r).encode() + b"\n" for r in reqs)
                 self._writer.write(payload)
                 await self._writer.drain()
+                t = (timeout_s if timeout_s is not None
+                     else self.query_timeout_s)
                 out = []
                 for _ in reqs:
-                    line = await asyncio.wait_for(
-                        self._reader.readline(),
-                        timeout_s if timeout_s is not None
-                        else self.query_timeout_s)
+                    line = await self._readline(t)
                     if not line:
                         raise WaldbError("connection closed by server")
-                    out.append(json.loads(line))
+                    out.append(json.loads(line.decode("utf-8")))
                 return out
             except (ConnectionError, OSError, asyncio.TimeoutError,
                     ValueError) as exc:
