@@ -67,6 +67,8 @@ class Wal:
         self._active_start = 0
         self.start = 0   # lowest LSN still on disk
         self.end = 0     # next append offset == current LSN
+        self._buf = bytearray()              # append_buffered frames
+        self._buf_at = 0                     # file LSN of _buf[0]
 
     # -------------------------------------------------------------- open
     def _seg_path(self, start: int) -> str:
@@ -140,29 +142,54 @@ class Wal:
 
     def close(self) -> None:
         if self._fd is not None:
+            self.flush_buffer()
             os.close(self._fd)
             self._fd = None
 
     # ------------------------------------------------------------ append
     def _maybe_roll(self) -> None:
         if self.end - self._active_start >= self.segment_bytes:
+            self.flush_buffer()
             os.fsync(self._fd)
             self._segs.append(self.end)
             with open(self._seg_path(self.end), "wb"):
                 pass
             self._open_active()
 
-    def append(self, payload: bytes) -> int:
-        """Append one record; returns its commit LSN."""
+    def flush_buffer(self) -> None:
+        """Write frames accumulated by ``append_buffered`` to the active
+        segment in ONE pwrite.  Every path that observes file contents
+        (read, fsync, roll, truncate, close) calls this first, so the
+        buffer is never visible as a gap."""
+        if self._buf:
+            os.pwrite(self._fd, bytes(self._buf),
+                      self._buf_at - self._active_start)
+            self._buf.clear()
+
+    def append_buffered(self, payload: bytes) -> int:
+        """``append()`` without the per-record pwrite: frames accumulate
+        in memory and hit the file at the next ``flush_buffer()``.  The
+        server flushes before any socket write and before fsync, and
+        replica reads flush implicitly (``read``), so observable
+        durability is unchanged — an ack never rests on bytes that are
+        only in this buffer."""
         self._maybe_roll()
         native = _codec()
         if native is not None:
             frame = native.encode_frame(payload)
         else:
             frame = _HDR.pack(len(payload), zlib.crc32(payload)) + payload
-        os.pwrite(self._fd, frame, self.end - self._active_start)
+        if not self._buf:
+            self._buf_at = self.end
+        self._buf += frame
         self.end += len(frame)
         return self.end
+
+    def append(self, payload: bytes) -> int:
+        """Append one record; returns its commit LSN."""
+        lsn = self.append_buffered(payload)
+        self.flush_buffer()
+        return lsn
 
     def append_raw(self, data: bytes, at: int) -> int:
         """Standby path: append raw replicated WAL bytes at offset ``at``
@@ -170,17 +197,20 @@ class Wal:
         if at != self.end:
             raise ValueError("non-contiguous WAL append (%d != %d)"
                              % (at, self.end))
+        self.flush_buffer()
         self._maybe_roll()
         os.pwrite(self._fd, data, self.end - self._active_start)
         self.end += len(data)
         return self.end
 
     def fsync(self) -> None:
+        self.flush_buffer()
         os.fsync(self._fd)
 
     # -------------------------------------------------------- truncation
     def truncate_to(self, lsn: int) -> None:
         """Timeline fencing: discard everything above ``lsn``."""
+        self.flush_buffer()
         keep = [s for s in self._segs if s <= lsn]
         drop = [s for s in self._segs if s > lsn]
         for s in drop:
@@ -219,6 +249,7 @@ class Wal:
         Raises ``WalGone`` if ``start`` predates the oldest segment."""
         if start < self.start:
             raise WalGone(start, self.start)
+        self.flush_buffer()
         n = min(self.end - start, max_bytes)
         if n <= 0:
             return b""
