@@ -423,6 +423,46 @@ def test_wal_class_segmentation_and_recycling(tmp_path):
     w4.close()
 
 
+def test_wal_buffered_append_flush_and_crash_drop(tmp_path):
+    """Direct Wal unit test for append_buffered: reads force a flush
+    mid-stream (replica senders observe everything appended), segment
+    rolls flush the old segment first, a crash before any flush drops
+    ONLY the still-buffered (never-observed, hence never-acked) tail,
+    and fsync makes buffered records fully durable."""
+    from manatee_amd.db.waldb.wal import Wal, encode_op
+
+    d = str(tmp_path / "wal")
+    w = Wal(d, segment_bytes=4096)
+    w.open()
+    lsns = [w.append_buffered(encode_op({"op": "put", "k": "k%d" % i,
+                                         "v": "x" * 100}))
+            for i in range(200)]           # crosses several rolls
+    assert len(w._segs) > 2
+    # a read (the replica-sender path) flushes implicitly
+    chunk = w.read_aligned(0)
+    assert chunk
+    for i in range(50):
+        w.append_buffered(encode_op({"op": "put", "k": "m%d" % i, "v": 1}))
+    end_before = w.end
+    # simulate kill -9 with the tail still buffered
+    w._buf.clear()
+    os.close(w._fd)
+    w._fd = None
+    w2 = Wal(d, segment_bytes=4096)
+    end = w2.open(replay=lambda lsn, p: None)
+    assert end >= lsns[-1], "flushed records must survive"
+    assert end <= end_before
+    # buffered records become durable through fsync
+    l = w2.append_buffered(encode_op({"op": "put", "k": "z", "v": 2}))
+    w2.fsync()
+    w2.close()
+    w3 = Wal(d, segment_bytes=4096)
+    seen = []
+    assert w3.open(replay=lambda lsn, p: seen.append(lsn)) == l
+    assert seen[-1] == l
+    w3.close()
+
+
 def test_checkpoint_bounds_recovery_and_wal_size(tmp_path):
     """A primary under write load checkpoints, recycles old WAL segments
     and recovers from checkpoint + tail after kill -9 with all data."""
