@@ -1,6 +1,7 @@
 """Write/read throughput micro-benchmark for the replicated engine.
 
     python -m manatee_amd.tools.wrbench [--seconds 10] [--writers 1,8,32]
+        [--engine waldb|postgres]
 
 Spins a local 3-peer shard (primary -> sync -> async) and measures,
 for each writer count, the rate of *synchronously acknowledged* puts
@@ -18,12 +19,11 @@ import sys
 import tempfile
 import time
 
-from ..db.waldb.client import WaldbClient
 from .devcluster import DevCluster
 
 
-async def writer_task(host, port, label, stop, counter, pipeline=0):
-    cli = WaldbClient(host, port)
+async def writer_task(peer, label, stop, counter, pipeline=0):
+    cli = peer.db_client()
     i = 0
     try:
         while not stop.is_set():
@@ -41,8 +41,8 @@ async def writer_task(host, port, label, stop, counter, pipeline=0):
         await cli.close()
 
 
-async def reader_task(host, port, stop, counter):
-    cli = WaldbClient(host, port)
+async def reader_task(peer, stop, counter):
+    cli = peer.db_client()
     i = 0
     try:
         while not stop.is_set():
@@ -62,11 +62,11 @@ async def measure(c: DevCluster, n_writers: int, seconds: float,
     wcount = [0]
     rcount = [0]
     tasks = [asyncio.ensure_future(
-        writer_task(prim.ip, prim.pg_port, "%d.%d" % (pipeline, w),
+        writer_task(prim, "%d.%d" % (pipeline, w),
                     stop, wcount, pipeline=pipeline))
         for w in range(n_writers)]
     tasks += [asyncio.ensure_future(
-        reader_task(sync.ip, sync.pg_port, stop, rcount))
+        reader_task(sync, stop, rcount))
         for _ in range(4)]
     t0 = time.monotonic()
     await asyncio.sleep(seconds)
@@ -78,8 +78,9 @@ async def measure(c: DevCluster, n_writers: int, seconds: float,
             "standby_reads_per_s": round(rcount[0] / dt, 1)}
 
 
-async def run(writers, seconds, workdir) -> dict:
-    c = DevCluster(workdir, n_peers=3, shard_name="1.wrbench")
+async def run(writers, seconds, workdir, engine="waldb") -> dict:
+    c = DevCluster(workdir, n_peers=3, shard_name="1.wrbench",
+                   engine=engine)
     try:
         await c.start()
         await c.wait_cluster(
@@ -103,7 +104,7 @@ async def run(writers, seconds, workdir) -> dict:
         cli = c.peer_by_id(s["primary"]["id"]).db_client()
         total = await cli.count(prefix="w-")
         await cli.close()
-        return {"results": results, "total_keys": total}
+        return {"engine": engine, "results": results, "total_keys": total}
     finally:
         c.stop()
 
@@ -111,13 +112,16 @@ async def run(writers, seconds, workdir) -> dict:
 def main(argv=None) -> int:
     ap = argparse.ArgumentParser(prog="manatee-wrbench")
     ap.add_argument("--seconds", type=float, default=10.0)
+    ap.add_argument("--engine", choices=("waldb", "postgres"),
+                    default="waldb")
     ap.add_argument("--writers", default="1,8,32")
     ap.add_argument("-d", "--dir", default=None)
     ns = ap.parse_args(argv)
     writers = [int(w) for w in ns.writers.split(",")]
     workdir = ns.dir or tempfile.mkdtemp(prefix="manatee-wrbench-")
     try:
-        out = asyncio.run(run(writers, ns.seconds, workdir))
+        out = asyncio.run(run(writers, ns.seconds, workdir,
+                              engine=ns.engine))
     finally:
         if ns.dir is None:
             shutil.rmtree(workdir, ignore_errors=True)
