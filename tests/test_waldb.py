@@ -144,6 +144,55 @@ def test_single_node_put_get_and_crash_recovery(tmp_path):
         n.stop()
 
 
+def test_repl_handoff_consumes_pipelined_ack(tmp_path):
+    """A standby's first ack arriving in the SAME TCP segment as its
+    repl request must not be lost: the server's chunked request framing
+    reads past the repl line, and the buffered leftover is handed to
+    the ack consumer (server._BufferedReader).  Without that handoff
+    the gated commit below would never be acknowledged."""
+    n = Node(tmp_path, "prim")
+    n.init()
+    n.write_conf(role="primary", sync_name="fakesync")
+    n.start()
+    try:
+        async def go():
+            c = n.client()
+            st = await c.status()
+            ident = st["ident"]
+            # this put appends immediately but blocks awaiting the sync
+            put_task = asyncio.ensure_future(c.put("gated", 1,
+                                                   timeout_s=20))
+            await asyncio.sleep(0.3)
+            assert not put_task.done()
+
+            st2c = n.client()
+            st2 = await st2c.status()
+            lsn = int(st2["current_lsn"].split("/")[1], 16)
+            await st2c.close()
+            assert lsn > 0
+
+            reader, writer = await asyncio.open_connection("127.0.0.1",
+                                                           n.port)
+            # repl request AND the ack for the appended record in ONE
+            # write -> one TCP segment -> one server-side read()
+            repl = json.dumps({"q": "repl", "name": "fakesync",
+                               "ident": ident, "timeline": 1,
+                               "start_lsn": 0})
+            ack = json.dumps({"write_lsn": lsn, "flush_lsn": lsn,
+                              "replay_lsn": lsn})
+            writer.write((repl + "\n" + ack + "\n").encode())
+            await writer.drain()
+            line = await asyncio.wait_for(reader.readline(), 10)
+            assert json.loads(line).get("ok") is True
+            # the pipelined ack must release the gated commit
+            await asyncio.wait_for(put_task, 10)
+            writer.close()
+            await c.close()
+        run(go())
+    finally:
+        n.stop()
+
+
 def test_sync_replication_gates_commit(tmp_path):
     prim = Node(tmp_path, "prim")
     sync = Node(tmp_path, "sync")
