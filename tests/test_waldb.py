@@ -193,6 +193,73 @@ def test_repl_handoff_consumes_pipelined_ack(tmp_path):
         n.stop()
 
 
+def test_reconnecting_replica_supersedes_stale_sender(tmp_path):
+    """A replica reconnect under the same name must EVICT the wedged
+    old connection immediately.  Before the fix, the stale row stayed
+    first in pg_stat_replication until wal_sender_timeout (default
+    60 s) reaped it, and the commit gate read its frozen ack LSNs —
+    the rare ~60 s failover outliers seen in long chaos soaks."""
+    n = Node(tmp_path, "prim")
+    n.init()
+    n.write_conf(role="primary", sync_name="fakesync")
+    n.start()
+    try:
+        async def go():
+            c = n.client()
+            st = await c.status()
+            ident = st["ident"]
+
+            async def connect_sync(ack_lsn):
+                reader, writer = await asyncio.open_connection(
+                    "127.0.0.1", n.port)
+                repl = json.dumps({"q": "repl", "name": "fakesync",
+                                   "ident": ident, "timeline": 1,
+                                   "start_lsn": 0})
+                ack = json.dumps({"write_lsn": ack_lsn,
+                                  "flush_lsn": ack_lsn,
+                                  "replay_lsn": ack_lsn})
+                writer.write((repl + "\n" + ack + "\n").encode())
+                await writer.drain()
+                line = await asyncio.wait_for(reader.readline(), 10)
+                assert json.loads(line).get("ok") is True
+                return reader, writer
+
+            # wedged "old" connection: acks LSN 0 and then goes silent
+            r1, w1 = await connect_sync(0)
+            await asyncio.sleep(0.2)
+            # a put now blocks on the silent sync
+            put_task = asyncio.ensure_future(c.put("stuck", 1,
+                                                   timeout_s=30))
+            await asyncio.sleep(0.3)
+            assert not put_task.done()
+            st2c = n.client()
+            lsn = int((await st2c.status())["current_lsn"].split("/")[1],
+                      16)
+            await st2c.close()
+
+            # the replica reconnects (same name) and acks the record:
+            # the commit must complete promptly — NOT after 60 s
+            t0 = time.monotonic()
+            r2, w2 = await connect_sync(lsn)
+            await asyncio.wait_for(put_task, 10)
+            assert time.monotonic() - t0 < 10
+            # exactly one replication row remains
+            st3c = n.client()
+            rows = (await st3c.status())["replication"]
+            assert len([r for r in rows
+                        if r["application_name"] == "fakesync"]) == 1
+            await st3c.close()
+            for w in (w1, w2):
+                try:
+                    w.close()
+                except Exception:
+                    pass
+            await c.close()
+        run(go())
+    finally:
+        n.stop()
+
+
 def test_sync_replication_gates_commit(tmp_path):
     prim = Node(tmp_path, "prim")
     sync = Node(tmp_path, "sync")
