@@ -98,11 +98,18 @@ def get_version_info(data_dir: str, data_conf: str,
     """{initialized, current} from manatee-config.json + PG_VERSION
     (ref getVersionInfo lib/postgresMgr.js:446-510)."""
     pgc = None
+    torn = False
     try:
         with open(data_conf) as f:
-            pgc = json.load(f)
+            raw = f.read()
+        if raw.strip():
+            pgc = json.loads(raw)
+        else:
+            torn = True
     except FileNotFoundError:
         pass
+    except ValueError:
+        torn = True
     curver = None
     try:
         with open(os.path.join(data_dir, "PG_VERSION")) as f:
@@ -111,6 +118,21 @@ def get_version_info(data_dir: str, data_conf: str,
         pass
 
     if pgc is None:
+        if torn:
+            # manatee-config.json exists but is empty/corrupt: a kill -9
+            # (or a snapshot/stream read) caught a pre-atomic-write
+            # rewrite mid-flight.  The dataset is modern (the file was
+            # created by this code), so reconstruct from PG_VERSION
+            # instead of failing every transition forever.
+            if curver is not None:
+                full = versions.get(curver)
+                if full is None:
+                    raise ValueError(
+                        "torn %s and no configured binaries for "
+                        "PG_VERSION %r" % (DATA_CONF_NAME, curver))
+                return {"initialized": full, "current": full}
+            full = versions[default_version]
+            return {"initialized": full, "current": full}
         if curver is None:
             full = versions[default_version]
             return {"initialized": full, "current": full}
@@ -176,8 +198,16 @@ class PostgresEngine(Engine):
         verinfo = get_version_info(self.data_dir, self.data_conf,
                                    self.versions, self.default_version)
         os.makedirs(self.dataset_dir, exist_ok=True)
-        with open(self.data_conf, "w") as f:
+        # atomic replace: this file is rewritten on EVERY transition and
+        # read by version resolution, snapshot copies and restore
+        # streams — a kill -9 mid-"w"-rewrite left it 0 bytes and every
+        # later transition failed (found by the 15-step postgres bench)
+        tmp = self.data_conf + ".tmp"
+        with open(tmp, "w") as f:
             json.dump(verinfo, f)
+            f.flush()
+            os.fsync(f.fileno())
+        os.replace(tmp, self.data_conf)
         version = verinfo["current"]
         self.current_version = version
         self.major = pg_strip_minor(version)

@@ -73,6 +73,52 @@ def test_version_info_cases(tmp_path):
         get_version_info(data, dconf, versions, "12")
 
 
+def test_version_info_torn_config_recovers(tmp_path):
+    """A 0-byte or half-written manatee-config.json (kill -9 caught the
+    pre-atomic rewrite, or a snapshot copied it mid-write) must be
+    reconstructed from PG_VERSION — it previously raised
+    JSONDecodeError on every transition, wedging the peer read-only
+    forever (found by the 15-step engine=postgres bench)."""
+    data = str(tmp_path / "data")
+    dconf = str(tmp_path / "manatee-config.json")
+    versions = {"9.6": "9.6.3", "12": "12.0"}
+    os.makedirs(data)
+    with open(os.path.join(data, "PG_VERSION"), "w") as f:
+        f.write("12\n")
+
+    for torn in ("", '{"initialized": "12.0", "cur'):
+        with open(dconf, "w") as f:
+            f.write(torn)
+        vi = get_version_info(data, dconf, versions, "12")
+        assert vi == {"initialized": "12.0", "current": "12.0"}
+
+    # torn file, no PG_VERSION either → default version
+    os.unlink(os.path.join(data, "PG_VERSION"))
+    with open(dconf, "w") as f:
+        f.write("")
+    vi = get_version_info(data, dconf, versions, "12")
+    assert vi == {"initialized": "12.0", "current": "12.0"}
+
+    # torn file with a PG_VERSION we have no binaries for stays fatal
+    with open(os.path.join(data, "PG_VERSION"), "w") as f:
+        f.write("11\n")
+    with pytest.raises(ValueError):
+        get_version_info(data, dconf, versions, "12")
+
+
+def test_resolve_versioned_paths_writes_config_atomically(tmp_path):
+    """resolve_versioned_paths must never leave manatee-config.json
+    observable in a truncated state: it writes tmp + fsync +
+    os.replace, so concurrent snapshot copies and kill -9 see either
+    the old or the new content."""
+    eng = mk_engine(tmp_path, "12")
+    eng.resolve_versioned_paths()
+    dconf = os.path.join(str(tmp_path / "store"), "manatee-config.json")
+    with open(dconf) as f:
+        assert json.load(f) == {"initialized": "12.0", "current": "12.0"}
+    assert not os.path.exists(dconf + ".tmp")
+
+
 def mk_engine(tmp_path, major, versions=None, extra_cfg=None):
     versions = versions or {"9.6": "9.6.3", "12": "12.0"}
     data = str(tmp_path / "store" / "data")
