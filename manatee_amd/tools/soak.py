@@ -139,6 +139,52 @@ async def verify(cluster: DevCluster, writer: SoakWriter) -> int:
         writer.resume()
 
 
+async def dump_stall(cluster: DevCluster, action: str) -> None:
+    """Diagnostics for a cycle that failed to become writable inside
+    the fault window (before the operator runbook runs): cluster state
+    plus every reachable db's own view.  The rare >60 s outliers in
+    long soaks are only debuggable from this moment's state."""
+    try:
+        s = await cluster.cluster_state()
+        print("# STALL[%s] cluster_state: %s"
+              % (action, json.dumps(s)[:700] if s else None),
+              file=sys.stderr)
+        for p in cluster.peers:
+            cli = p.db_client()
+            try:
+                st = await asyncio.wait_for(cli.status(), 2.0)
+                print("# STALL[%s] db %s: role=%s ro=%s sync=%s up=%s "
+                      "cur=%s repl=%s"
+                      % (action, p.id, st.get("role"),
+                         st.get("read_only"), st.get("sync_standby"),
+                         st.get("upstream_status"),
+                         st.get("current_lsn"),
+                         json.dumps(st.get("replication"))[:300]),
+                      file=sys.stderr)
+            except Exception as exc:
+                print("# STALL[%s] db %s: unreachable (%r)"
+                      % (action, p.id, exc), file=sys.stderr)
+            finally:
+                try:
+                    await cli.close()
+                except Exception:
+                    pass
+            try:
+                _code, body = await asyncio.wait_for(
+                    p.http_status("/state"), 2.0)
+                st = json.loads(body)
+                print("# STALL[%s] sitter %s: role=%s dbOnline=%s gen=%s"
+                      % (action, p.id, st.get("role"), st.get("dbOnline"),
+                         (st.get("clusterState") or {}).get("generation")),
+                      file=sys.stderr)
+            except Exception as exc:
+                print("# STALL[%s] sitter %s: unreachable (%r)"
+                      % (action, p.id, exc), file=sys.stderr)
+    except Exception as exc:
+        print("# STALL[%s] dump failed: %r" % (action, exc),
+              file=sys.stderr)
+
+
 ACTIONS = ["kill_primary", "kill_sync", "kill_async",
            "kill_db_only", "pause_primary", "zk_outage",
            "partition_zk_primary", "partition_repl", "partition_asym"]
@@ -234,6 +280,7 @@ async def soak(minutes: float, seed: int, workdir: str,
             try:
                 await c.wait_writable(timeout_s=60)
             except AssertionError:
+                await dump_stall(c, action)
                 # legitimately unavailable states exist (e.g. primary
                 # dead while the only other healthy peer is the sync and
                 # the third is deposed) — run the operator runbook:
