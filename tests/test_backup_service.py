@@ -140,3 +140,63 @@ def test_stalled_receiver_fails_job_without_wedging_queue(tmp_path):
             ok_srv.close()
             await sender.stop()
     run(go())
+
+
+def test_restore_fails_fast_when_source_freezes_mid_stream(tmp_path):
+    """A restore whose SOURCE peer stops responding mid-stream (SIGSTOP,
+    partition) must fail within ~one job-poll timeout so the FSM can
+    re-evaluate — not hang for the full restore deadline.  This is the
+    guard that keeps a takeover candidate from being wedged by a frozen
+    restore source."""
+    import json as _json
+    import time as _time
+
+    from manatee_amd.backup.restore import RestoreClient, RestoreError
+
+    async def go():
+        st = DirStore(str(tmp_path / "dst"), log=null_logger())
+
+        # a minimal "backup source" that accepts the job, streams a few
+        # bytes to the restore listener, then freezes: the job-status
+        # poll connection is accepted but NEVER answered
+        async def handle(reader, writer):
+            req = await reader.readuntil(b"\r\n\r\n")
+            line = req.split(b"\r\n", 1)[0].decode()
+            if line.startswith("POST /backup"):
+                hdrs = req.decode()
+                n = int([h for h in hdrs.split("\r\n")
+                         if h.lower().startswith("content-length")]
+                        [0].split(":")[1])
+                body = _json.loads((await reader.readexactly(n)).decode())
+                resp = _json.dumps({"jobid": "j1",
+                                    "jobPath": "/backup/j1"}).encode()
+                writer.write(b"HTTP/1.1 200 OK\r\nContent-Type: "
+                             b"application/json\r\nContent-Length: %d"
+                             b"\r\n\r\n%s" % (len(resp), resp))
+                await writer.drain()
+
+                async def stream():
+                    r2, w2 = await asyncio.open_connection(
+                        body["host"], body["port"])
+                    w2.write(b"partial-tar-bytes")
+                    await w2.drain()
+                    # ... then freeze: never close, never send more
+                    await asyncio.sleep(3600)
+                asyncio.ensure_future(stream())
+            else:
+                # job-status poll: accepted, never answered (frozen)
+                await asyncio.sleep(3600)
+
+        src = await asyncio.start_server(handle, "127.0.0.1", 0)
+        src_port = src.sockets[0].getsockname()[1]
+        rc = RestoreClient(st, "127.0.0.1", poll_interval_s=0.2,
+                           log=null_logger())
+        t0 = _time.monotonic()
+        with pytest.raises(Exception):
+            await rc.restore("http://127.0.0.1:%d" % src_port,
+                             timeout_s=300.0, isolate=False)
+        dt = _time.monotonic() - t0
+        assert dt < 30, "restore hung %.1fs against a frozen source" % dt
+        assert rc.restore_object.failed
+        src.close()
+    run(go(), timeout=90)
