@@ -149,11 +149,26 @@ async def kill_escalate(pid: int, ops_timeout_s: float,
         except (OSError, IndexError):
             return False
 
+    def stopped() -> bool:
+        try:
+            with open("/proc/%d/stat" % pid, "r") as f:
+                stat = f.read()
+            return stat.rpartition(")")[2].split()[0] == "T"
+        except (OSError, IndexError):
+            return False
+
     for sig in signals:
         if not alive():
             return 0
         try:
             os.kill(target, sig)
+            if sig != signal.SIGKILL and stopped():
+                # a SIGSTOPped child QUEUES termination signals but
+                # cannot run them — without a CONT the escalation
+                # would burn its full per-step timeout before SIGKILL
+                # (observed as rare ~60 s failovers when chaos froze a
+                # db out from under a live sitter); wake it to die
+                os.kill(target, signal.SIGCONT)
         except ProcessLookupError:
             return 0
         deadline = time.monotonic() + ops_timeout_s

@@ -236,37 +236,59 @@ class DevPeer:
         self.backup_proc = None
         self.snap_proc = None
 
+    @staticmethod
+    def _signal_pids(pids, sig) -> None:
+        for pid in pids:
+            try:
+                os.killpg(pid, sig)
+            except (ProcessLookupError, PermissionError):
+                try:
+                    os.kill(pid, sig)
+                except ProcessLookupError:
+                    pass
+
+    @staticmethod
+    def _proc_state(pid: int) -> str:
+        try:
+            with open("/proc/%d/stat" % pid) as f:
+                return f.read().rpartition(")")[2].split()[0]
+        except (OSError, IndexError):
+            return "?"
+
     def pause(self) -> None:
         """SIGSTOP the whole peer (sitter pg + db pg) — the network-
         partition analogue on one host: processes stay alive but stop
         responding, the ZK session expires, and on resume() the peer
-        discovers the cluster moved on without it."""
-        pids = [p.pid for p in (self.sitter_proc, self.backup_proc,
-                                self.snap_proc)
-                if p is not None and p.poll() is None]
-        pids += self.db_pids()
-        for pid in pids:
-            try:
-                os.killpg(pid, signal.SIGSTOP)
-            except (ProcessLookupError, PermissionError):
-                try:
-                    os.kill(pid, signal.SIGSTOP)
-                except ProcessLookupError:
-                    pass
+        discovers the cluster moved on without it.
+
+        The sitter is frozen FIRST and its stop confirmed before the
+        db pids are read: a live sitter can respawn the db at any
+        moment, and a half-frozen peer (db stopped, sitter running)
+        produces misleading chaos results.  The stopped pids are
+        recorded so resume() wakes exactly what was frozen even if pid
+        files changed meanwhile."""
+        daemons = [p.pid for p in (self.sitter_proc, self.backup_proc,
+                                   self.snap_proc)
+                   if p is not None and p.poll() is None]
+        self._signal_pids(daemons, signal.SIGSTOP)
+        deadline = time.monotonic() + 2.0
+        while time.monotonic() < deadline:
+            if all(self._proc_state(pid) in ("T", "Z", "?")
+                   for pid in daemons):
+                break
+            time.sleep(0.01)
+        dbs = self.db_pids()
+        self._signal_pids(dbs, signal.SIGSTOP)
+        self._paused_pids = daemons + dbs
 
     def resume(self) -> None:
-        pids = [p.pid for p in (self.sitter_proc, self.backup_proc,
-                                self.snap_proc)
-                if p is not None and p.poll() is None]
-        pids += self.db_pids()
-        for pid in pids:
-            try:
-                os.killpg(pid, signal.SIGCONT)
-            except (ProcessLookupError, PermissionError):
-                try:
-                    os.kill(pid, signal.SIGCONT)
-                except ProcessLookupError:
-                    pass
+        pids = list(getattr(self, "_paused_pids", []))
+        self._paused_pids = []
+        for p in (self.sitter_proc, self.backup_proc, self.snap_proc):
+            if p is not None and p.poll() is None and p.pid not in pids:
+                pids.append(p.pid)
+        pids += [pid for pid in self.db_pids() if pid not in pids]
+        self._signal_pids(pids, signal.SIGCONT)
 
     def kill_db_only(self) -> None:
         """SIGKILL only the database child (the sitter must notice and

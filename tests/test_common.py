@@ -178,6 +178,34 @@ def test_kill_escalate():
         proc.wait(timeout=10)
 
 
+def test_kill_escalate_wakes_stopped_target():
+    """Escalating against a SIGSTOPped child must not burn a full
+    per-step timeout waiting for queued SIGINT/SIGQUIT it can never
+    run: the escalation CONTs a stopped target so it can die.  Without
+    the CONT this took 2 x ops_timeout before SIGKILL — the rare ~60 s
+    failover outliers when chaos froze a db under a live sitter."""
+    import asyncio
+    import subprocess
+    import time as _time
+
+    proc = subprocess.Popen(
+        ["/usr/bin/python3", "-c",
+         "import time; print('ready', flush=True); time.sleep(120)"],
+        stdout=subprocess.PIPE)
+    assert proc.stdout.readline().strip() == b"ready"
+    os.kill(proc.pid, signal.SIGSTOP)
+    try:
+        t0 = _time.monotonic()
+        sig = asyncio.run(procutil.kill_escalate(proc.pid,
+                                                 ops_timeout_s=20.0))
+        dt = _time.monotonic() - t0
+        # SIGINT + CONT kills a default-disposition python promptly
+        assert sig == signal.SIGINT
+        assert dt < 5, "stopped child burned %.1fs of escalation" % dt
+    finally:
+        proc.wait(timeout=10)
+
+
 def test_run_async():
     import asyncio
 
