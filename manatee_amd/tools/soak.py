@@ -263,6 +263,7 @@ async def soak(minutes: float, seed: int, workdir: str,
                     stats.setdefault("slow_zk_takeovers", 0)
                     stats["slow_zk_takeovers"] += 1
                     print("# WARN: %s" % exc, file=sys.stderr)
+                    await dump_stall(c, action)
                 finally:
                     c.heal_zk(prim)
             elif action == "partition_repl":
