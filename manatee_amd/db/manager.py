@@ -46,6 +46,25 @@ from ..storage.provider import SnapshotStore
 from .engine import Engine, peer_id_from_urls
 
 
+def db_child_preexec() -> None:
+    """preexec_fn for the database child.
+
+    SIGHUP is ignored until the db installs its own reload handler (the
+    default disposition would kill the interpreter during boot).
+
+    SIGINT/SIGQUIT/SIGTERM are reset to SIG_DFL: the spawning chain may
+    have them at SIG_IGN (POSIX sets exactly that for commands
+    backgrounded without job control), and ignored dispositions survive
+    exec — an inheriting db child would be unkillable-except-SIGKILL
+    until its event loop installs handlers, turning a dirty stop that
+    lands in the boot/recovery window into a 2 x ops_timeout (= 60 s)
+    stall that blocks the serialized FSM mid-failover (found by long
+    chaos soaks as rare ~60 s failover outliers)."""
+    signal.signal(signal.SIGHUP, signal.SIG_IGN)
+    for _s in (signal.SIGINT, signal.SIGQUIT, signal.SIGTERM):
+        signal.signal(_s, signal.SIG_DFL)
+
+
 class DbManager:
     def __init__(self, *, engine: Engine, store: SnapshotStore,
                  ip: str,
@@ -156,16 +175,10 @@ class DbManager:
         self._expect_exit = False
         argv = self.engine.spawn_argv()
         self.log.info("starting database", argv=argv)
-        def _child_init():
-            # the db installs its own SIGHUP handler once its event loop
-            # is up; until then a reload signal must be ignored, not
-            # fatal (default SIGHUP disposition kills the interpreter)
-            signal.signal(signal.SIGHUP, signal.SIG_IGN)
-
         self._proc = await asyncio.create_subprocess_exec(
             *argv, stdout=asyncio.subprocess.DEVNULL,
             stderr=asyncio.subprocess.DEVNULL,
-            start_new_session=True, preexec_fn=_child_init)
+            start_new_session=True, preexec_fn=db_child_preexec)
         # record the child pid IMMEDIATELY: the db writes its own pid file
         # only once it is up, and anything that needs to SIGKILL the whole
         # peer (tests, operators) must not race that window.  Written both

@@ -157,11 +157,21 @@ class DevPeer:
             env["MANATEE_DIAL_MAP"] = os.path.join(self.dir,
                                                    "dialmap.json")
         logf = open(os.path.join(self.dir, logname), "a")
+
+        def _reset_signals():
+            # a backgrounded test runner (nohup … &) has SIGINT/SIGQUIT
+            # set to SIG_IGN, which survives exec — the daemons (and
+            # anything they spawn) must get default dispositions or
+            # dirty-stop escalation cannot touch them during boot
+            for s in (signal.SIGINT, signal.SIGQUIT, signal.SIGTERM):
+                signal.signal(s, signal.SIG_DFL)
+
         return subprocess.Popen(
             [sys.executable, "-m", module, "-f",
              os.path.join(self.dir, config), "-v",
              "--log-file", os.path.join(self.dir, logname + ".json")],
-            env=env, stdout=logf, stderr=logf, start_new_session=True)
+            env=env, stdout=logf, stderr=logf, start_new_session=True,
+            preexec_fn=_reset_signals)
 
     def start(self) -> None:
         self.write_configs()

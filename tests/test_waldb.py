@@ -144,6 +144,52 @@ def test_single_node_put_get_and_crash_recovery(tmp_path):
         n.stop()
 
 
+def test_db_child_killable_during_boot_despite_inherited_sigign(tmp_path):
+    """A db child spawned from a chain with SIGINT/SIGQUIT at SIG_IGN
+    (POSIX backgrounding does this; dispositions survive exec) must
+    still die promptly when a dirty stop lands in its BOOT window —
+    before the event loop installs handlers.  Without the preexec
+    reset, SIGINT and SIGQUIT were silently ignored for the whole boot
+    and escalation burned 2 x ops_timeout to SIGKILL, stalling the
+    serialized FSM ~60 s mid-failover (found by long chaos soaks)."""
+    import subprocess
+
+    from manatee_amd.db.manager import db_child_preexec
+
+    data = str(tmp_path / "db")
+    init_data_dir(data)
+    confparser.write(os.path.join(data, "waldb.conf"), {
+        "role": "primary", "listen_ip": "127.0.0.1", "port": "0",
+        "name": "n1"})
+
+    old_int = signal.signal(signal.SIGINT, signal.SIG_IGN)
+    old_quit = signal.signal(signal.SIGQUIT, signal.SIG_IGN)
+    try:
+        env = dict(os.environ, PYTHONPATH=REPO)
+        proc = subprocess.Popen(
+            [sys.executable, "-m", "manatee_amd.db.waldb.server",
+             "-D", data], env=env, start_new_session=True,
+            stdout=subprocess.DEVNULL, stderr=subprocess.DEVNULL,
+            preexec_fn=db_child_preexec)
+        try:
+            # SIGINT immediately — almost certainly inside the boot
+            # window (python imports + recovery), before any handler
+            time.sleep(0.05)
+            os.killpg(proc.pid, signal.SIGINT)
+            deadline = time.monotonic() + 8
+            while proc.poll() is None and time.monotonic() < deadline:
+                time.sleep(0.05)
+            assert proc.poll() is not None, \
+                "db child ignored SIGINT during boot"
+        finally:
+            if proc.poll() is None:
+                os.killpg(proc.pid, signal.SIGKILL)
+                proc.wait()
+    finally:
+        signal.signal(signal.SIGINT, old_int)
+        signal.signal(signal.SIGQUIT, old_quit)
+
+
 def test_repl_handoff_consumes_pipelined_ack(tmp_path):
     """A standby's first ack arriving in the SAME TCP segment as its
     repl request must not be lost: the server's chunked request framing
