@@ -82,12 +82,16 @@ class ZkClient:
     async def close(self) -> None:
         self._closing = True
         if self.state == "connected" and self._writer is not None:
-            try:
+            async def _send_close():
                 async with self._send_lock:
                     w = jute.encode_request_header(self._next_xid(),
                                                    jute.OP_CLOSE_SESSION)
                     self._writer.write(w.framed())
                     await self._writer.drain()
+            try:
+                # bounded: a wedged server/socket must not hang close()
+                # (the session expires server-side anyway)
+                await asyncio.wait_for(_send_close(), 5.0)
             except Exception:
                 pass
         for task in (self._mgr_task, self._io_task, self._ping_task):
