@@ -172,9 +172,11 @@ async def dump_stall(cluster: DevCluster, action: str) -> None:
             try:
                 _code, body = await asyncio.wait_for(
                     p.http_status("/state"), 2.0)
-                st = json.loads(body)
-                print("# STALL[%s] sitter %s: role=%s dbOnline=%s gen=%s"
-                      % (action, p.id, st.get("role"), st.get("dbOnline"),
+                st = body if isinstance(body, dict) else json.loads(body)
+                print("# STALL[%s] sitter %s: role=%s state=%s "
+                      "dbOnline=%s gen=%s"
+                      % (action, p.id, st.get("role"),
+                         st.get("peerState"), st.get("dbOnline"),
                          (st.get("clusterState") or {}).get("generation")),
                       file=sys.stderr)
             except Exception as exc:
