@@ -260,16 +260,25 @@ class DbManager:
             await self.engine.close()
         except Exception:
             pass
+        t0 = time.monotonic()
+        sig_used = 0
         if proc.returncode is None:
             try:
-                await procutil.kill_escalate(proc.pid, self.ops_timeout_s,
-                                             pgid=True)
+                sig_used = await procutil.kill_escalate(
+                    proc.pid, self.ops_timeout_s, pgid=True)
             except (procutil.ExecError, ProcessLookupError):
                 pass
+        t_kill = time.monotonic() - t0
         try:
             await asyncio.wait_for(proc.wait(), self.ops_timeout_s)
         except asyncio.TimeoutError:
             pass
+        if t_kill > 2.0:
+            # a dirty stop should land on the FIRST signal in
+            # milliseconds — anything slower stalls the serialized FSM
+            # and deserves a trace (the ~60 s failover-outlier class)
+            self.log.warn("slow database stop", seconds=round(t_kill, 2),
+                          signal=sig_used, pid=proc.pid)
         self.log.info("database stopped")
 
     async def _restart_db(self) -> None:
