@@ -190,6 +190,40 @@ def test_db_child_killable_during_boot_despite_inherited_sigign(tmp_path):
         signal.signal(signal.SIGQUIT, old_quit)
 
 
+def test_dirty_exit_is_immediate_even_mid_checkpoint(tmp_path):
+    """A stop signal must end the process in milliseconds regardless of
+    what is in flight (MANATEE-188: durability comes from the WAL, not
+    a clean shutdown).  Before the os._exit change, the graceful
+    asyncio unwind WAITED for the default executor, where a whole-kv
+    checkpoint write could run for tens of seconds under load — the
+    dirty-stop escalation then burned 2 x ops_timeout to SIGKILL,
+    the dominant cause of rare ~60 s failovers in long chaos soaks."""
+    n = Node(tmp_path, "n1")
+    n.init()
+    # tiny checkpoint threshold: the server checkpoints continuously
+    # under this load, so a SIGINT lands mid-checkpoint with high
+    # probability
+    n.write_conf(role="primary", extra={"checkpoint_wal_bytes": "65536"})
+    n.start()
+    try:
+        async def load():
+            cli = n.client()
+            for i in range(40):
+                await cli.put_many((("ck-%d-%d" % (i, j), "v" * 200)
+                                    for j in range(500)))
+            await cli.close()
+        run(load())
+        t0 = time.monotonic()
+        os.kill(n.proc.pid, signal.SIGINT)
+        while n.proc.poll() is None and time.monotonic() - t0 < 10:
+            time.sleep(0.02)
+        dt = time.monotonic() - t0
+        assert n.proc.poll() is not None, "server survived SIGINT 10s"
+        assert dt < 3.0, "dirty exit took %.2fs" % dt
+    finally:
+        n.stop()
+
+
 def test_repl_handoff_consumes_pipelined_ack(tmp_path):
     """A standby's first ack arriving in the SAME TCP segment as its
     repl request must not be lost: the server's chunked request framing
