@@ -1103,6 +1103,54 @@ def test_sync_ack_pins_wal_retention_across_disconnect(tmp_path):
     run(go())
 
 
+def test_checkpoint_spacing_scales_with_image_size(tmp_path):
+    """Checkpoint I/O must stay amortized O(1) per WAL byte: once an
+    image has been written, the next checkpoint is due only after at
+    least half that image's size in NEW WAL (never sooner than the
+    configured checkpoint_wal_bytes).  Without this, a multi-hundred-MB
+    store re-serialized its entire kv every 8 MB of appends — the I/O
+    storm behind the shutdown-wait failover outliers."""
+    async def go():
+        from manatee_amd.common.logging import null_logger
+        from manatee_amd.db.waldb.server import WaldbServer
+
+        data = str(tmp_path / "n1")
+        init_data_dir(data)
+        confparser.write(os.path.join(data, "waldb.conf"), {
+            "role": "primary", "listen_ip": "127.0.0.1", "port": "0",
+            "name": "n1", "checkpoint_wal_bytes": "4096"})
+        srv = WaldbServer(data, null_logger())
+        await srv.start()
+        try:
+            for i in range(60):
+                await srv._do_write({"op": "put", "k": "k%d" % i,
+                                     "v": "x" * 100})
+            await srv._checkpoint()
+            first_lsn = srv.ckpt_lsn
+            assert srv._last_ckpt_bytes > 4096, \
+                "image should exceed the configured threshold"
+            # append a bit more than checkpoint_wal_bytes but LESS than
+            # half the image: the flusher must NOT schedule a new one
+            for i in range(10):
+                await srv._do_write({"op": "put", "k": "s%d" % i,
+                                     "v": "y" * 100})
+            due = max(srv.ckpt_wal_bytes, srv._last_ckpt_bytes // 2)
+            assert srv.replay_lsn - srv.ckpt_lsn < due
+            # ...but once past the adaptive threshold, it is due again
+            while srv.replay_lsn - srv.ckpt_lsn < due:
+                await srv._do_write({"op": "put", "k": "t%d" % srv.wal.end,
+                                     "v": "z" * 200})
+            await srv._checkpoint()
+            assert srv.ckpt_lsn > first_lsn
+        finally:
+            if srv._server is not None:
+                srv._server.close()
+            if srv._flusher is not None:
+                srv._flusher.cancel()
+            srv.wal.close()
+    run(go())
+
+
 def test_checkpoint_fsyncs_wal_before_publishing(tmp_path):
     """Regression (advisor finding): the checkpoint must fsync the WAL
     BEFORE capturing/publishing its LSN — a checkpoint.lsn beyond the
