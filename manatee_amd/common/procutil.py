@@ -175,6 +175,14 @@ async def kill_escalate(pid: int, ops_timeout_s: float,
         while time.monotonic() < deadline:
             if not alive():
                 return sig
+            if sig != signal.SIGKILL and stopped():
+                # the child may stop AFTER the signal was sent (or the
+                # stop may only become visible now); same rationale as
+                # above — wake it so the queued signal can run
+                try:
+                    os.kill(target, signal.SIGCONT)
+                except ProcessLookupError:
+                    return sig
             await asyncio.sleep(poll_s)
     if alive():
         raise ExecError(["kill", str(pid)], None, "", "",

@@ -194,6 +194,14 @@ def test_kill_escalate_wakes_stopped_target():
         stdout=subprocess.PIPE)
     assert proc.stdout.readline().strip() == b"ready"
     os.kill(proc.pid, signal.SIGSTOP)
+    # wait for the stop to be VISIBLE (state T) so the escalation's
+    # stopped-target handling — not a send/stop race — is what's tested
+    deadline = _time.monotonic() + 10
+    while _time.monotonic() < deadline:
+        with open("/proc/%d/stat" % proc.pid) as f:
+            if f.read().rpartition(")")[2].split()[0] == "T":
+                break
+        _time.sleep(0.01)
     try:
         t0 = _time.monotonic()
         sig = asyncio.run(procutil.kill_escalate(proc.pid,
