@@ -147,7 +147,17 @@ class ZkClient:
             # block until the io task dies (connection lost)
             try:
                 await asyncio.shield(self._io_task)
-            except (asyncio.CancelledError, Exception):
+            except asyncio.CancelledError:
+                # either THIS task was cancelled (teardown that never
+                # called close(), e.g. an event loop shutting down) or
+                # the io task was (close() is the only caller): stop
+                # managing.  Swallowing this and looping on made the
+                # manager task survive cancellation and reconnect
+                # forever — asyncio.run's final task-gather then hung a
+                # whole pytest process for its remaining per-test budget
+                self._teardown_conn()
+                raise
+            except Exception:
                 pass
             if self._closing:
                 return

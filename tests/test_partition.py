@@ -30,7 +30,37 @@ from manatee_amd.tools.devcluster import DevCluster  # noqa: E402
 
 
 def run(coro, timeout=240):
-    return asyncio.run(asyncio.wait_for(coro, timeout))
+    """asyncio.run with BOUNDED teardown.  asyncio.run's own
+    _cancel_all_tasks sends ONE cancel and gathers without a timeout; a
+    straggler task that blocks in a finally-await after that cancel
+    hangs the whole pytest process (observed: this file's pg-engine
+    test wedging at the 600 s pytest timeout AFTER the test body
+    finished).  Cancel leftovers here with a grace period, dump any
+    task that survives, and cancel it again so the runner's own pass
+    finds nothing left to wait on."""
+    import traceback
+
+    async def _main():
+        try:
+            return await asyncio.wait_for(coro, timeout)
+        finally:
+            cur = asyncio.current_task()
+            stragglers = [t for t in asyncio.all_tasks()
+                          if t is not cur and not t.done()]
+            for t in stragglers:
+                t.cancel()
+            if stragglers:
+                _done, pending = await asyncio.wait(stragglers, timeout=15)
+                for t in pending:
+                    print("WEDGED TASK (survived cancel):", t,
+                          file=sys.stderr)
+                    for f in t.get_stack():
+                        traceback.print_stack(f, file=sys.stderr)
+                    t.cancel()
+                if pending:
+                    await asyncio.wait(pending, timeout=10)
+
+    return asyncio.run(_main())
 
 
 async def _formed(cluster_dir, shard):

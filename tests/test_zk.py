@@ -322,3 +322,26 @@ def test_ensemble_conn_string_skips_dead_servers():
         await cli.close()
         await srv.stop()
     run(go())
+
+
+def test_connection_manager_dies_on_bare_cancel():
+    """asyncio.run's shutdown path cancels every leftover task ONCE and
+    gathers them with no timeout.  The connection manager must
+    therefore terminate on a bare cancel even mid-session (without
+    close() having set _closing): it used to swallow the CancelledError
+    raised through the io-task shield and loop back into reconnecting,
+    surviving the cancel forever and wedging interpreter shutdown —
+    observed as a whole pytest process hanging AFTER a partition test
+    had already passed."""
+    async def go():
+        srv, cli = await _pair()
+        try:
+            assert cli.state == "connected"
+            cli._mgr_task.cancel()
+            done, pending = await asyncio.wait({cli._mgr_task}, timeout=5)
+            assert not pending, \
+                "connection manager survived a bare cancel"
+        finally:
+            await cli.close()
+            await srv.stop()
+    run(go())
