@@ -120,11 +120,40 @@ async def _deposed_acks_must_drain(old_prim, new_prim, grace_s=4.0):
             got = await ncli.get(key)
         finally:
             await ncli.close()
-        assert got == "x", \
-            "old primary acknowledged a write ABSENT from the new " \
-            "primary: acked-write loss / split brain"
-        assert time.monotonic() < deadline, \
-            "old primary still acknowledging writes after the grace window"
+        async def _st(peer):
+            c2 = peer.db_client()
+            out = {}
+            try:
+                if hasattr(c2, "query"):        # waldb json protocol
+                    return await c2.query({"q": "status"}, timeout_s=3.0)
+                for label, sql in (
+                        ("stat_replication",
+                         "SELECT * FROM pg_stat_replication;"),
+                        ("wal_receiver",
+                         "SELECT * FROM pg_stat_wal_receiver;"),
+                        ("in_recovery", "SELECT pg_is_in_recovery();"),
+                        ("wal_lsn", "SELECT pg_current_wal_lsn();")):
+                    try:
+                        r = await c2._query(sql, timeout_s=3.0)
+                        out[label] = getattr(r, "rows", r)
+                    except Exception as exc:
+                        out[label] = repr(exc)
+                return out
+            except Exception as exc:
+                return {"error": repr(exc)}
+            finally:
+                await c2.close()
+        if got != "x":
+            raise AssertionError(
+                "old primary acknowledged a write ABSENT from the new "
+                "primary: acked-write loss / split brain\n"
+                "old_prim status: %r\nnew_prim status: %r"
+                % (await _st(old_prim), await _st(new_prim)))
+        if time.monotonic() >= deadline:
+            raise AssertionError(
+                "old primary still acknowledging writes after the grace "
+                "window\nold_prim status: %r\nnew_prim status: %r"
+                % (await _st(old_prim), await _st(new_prim)))
 
 
 def test_primary_cut_from_zk_but_not_from_clients(tmp_path):

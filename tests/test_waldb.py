@@ -1260,3 +1260,84 @@ def test_group_commit_pipeline_order_and_gate_break(tmp_path):
     finally:
         prim.stop()
         sync.stop()
+
+
+def test_promote_severs_repl_even_if_cancel_is_lost(tmp_path, monkeypatch):
+    """Python 3.10's asyncio.wait_for can LOSE a task cancellation that
+    races the inner read's completion (bpo-42130, fixed in 3.12) — and
+    under streaming load the replication client lives inside
+    wait_for(readexactly(...)).  A promote whose repl-task cancel was
+    eaten left the promoted sync streaming from AND ACKING the deposed
+    primary, whose sync-commit gate those acks kept open: it went on
+    acknowledging writes the new timeline will never contain (split
+    brain, observed ~1/6 under CPU load in the zk-partition tests).
+    Promote must therefore sever the link without relying on the
+    cancel: generation-bump + transport abort.  The WALDB_TEST_
+    EAT_REPL_CANCEL seam simulates the lost cancel deterministically;
+    on the pre-fix code this test fails (old primary keeps acking)."""
+    import shutil
+    prim = Node(tmp_path, "prim")
+    sync = Node(tmp_path, "sync")
+    prim.init()
+    prim.write_conf(role="primary", sync_name="sync")
+    prim.start()
+    monkeypatch.setenv("WALDB_TEST_EAT_REPL_CANCEL", "1")
+    try:
+        shutil.copytree(prim.data_dir, sync.data_dir,
+                        ignore=shutil.ignore_patterns("waldb.pid",
+                                                      "waldb.conf"))
+        sync.write_conf(role="standby",
+                        upstream="127.0.0.1:%d" % prim.port)
+        sync.start()
+
+        async def go():
+            c = prim.client()
+            sc = sync.client()
+
+            async def streaming():
+                st = await sc.status()
+                return st["upstream_status"] == "streaming"
+            await wait_async(streaming, what="sync streaming")
+            await c.put("before", 1, timeout_s=5.0)
+
+            # promote the sync (conf role flip + trigger + SIGHUP),
+            # with its repl-task cancel suppressed by the seam
+            sync.write_conf(role="primary")
+            sync.promote_trigger()
+            sync.sighup()
+
+            async def promoted():
+                st = await sc.status()
+                return st["role"] == "primary" and st["timeline"] == 2
+            await wait_async(promoted, what="sync promoted")
+
+            # the deposed primary's gate must close: within a short
+            # window no further write may be acknowledged (the severed
+            # link can carry no more acks)
+            deadline = time.monotonic() + 6.0
+            while True:
+                try:
+                    await c.put("after-%f" % time.monotonic(), 1,
+                                timeout_s=1.0)
+                    acked = True
+                except WaldbError:
+                    acked = False
+                if not acked:
+                    break
+                assert time.monotonic() < deadline, \
+                    "deposed primary still acknowledging writes after " \
+                    "promote: the replication link survived"
+            # the aborted socket surfaces at the sender's next keepalive
+            # (≤5 s): the replica row must then disappear — proof the
+            # link is dead at the TCP level, not merely quiet
+            async def replica_gone():
+                st = await c.status()
+                return st["replication"] == []
+            await wait_async(replica_gone, timeout=12.0,
+                             what="deposed primary drops the dead sender")
+            await c.close()
+            await sc.close()
+        run(go())
+    finally:
+        prim.stop()
+        sync.stop()
