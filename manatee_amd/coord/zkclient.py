@@ -98,7 +98,15 @@ class ZkClient:
             if task is not None:
                 task.cancel()
                 try:
-                    await task
+                    # bounded: close() sits on the failover-critical
+                    # session-rebuild path; a task whose cancel was
+                    # lost (bpo-42130-style) must not wedge it
+                    _done, pending = await asyncio.wait({task},
+                                                        timeout=5.0)
+                    for t in pending:
+                        t.cancel()
+                    if pending:
+                        await asyncio.wait(pending, timeout=2.0)
                 except (asyncio.CancelledError, Exception):
                     pass
         self._teardown_conn()
